@@ -43,11 +43,23 @@ import torch.nn as nn
 from .comm import Communicator, Work
 
 
+def _align8(n: int) -> int:
+    """Slot offsets are 8-element aligned so every slot is >=16B-aligned
+    for the vectorized HIP pack/unpack kernels (pad slots reduce as
+    zeros — harmless)."""
+    return (n + 7) & ~7
+
+
 class _Bucket:
     def __init__(self, params: List[nn.Parameter], dtype: torch.dtype,
                  device: torch.device, comm_dtype: torch.dtype):
         self.params = params
-        self.numel = sum(p.numel() for p in params)
+        self.offsets: Dict[nn.Parameter, int] = {}
+        off = 0
+        for p in params:
+            self.offsets[p] = off
+            off += _align8(p.numel())
+        self.numel = off
         self.flat = torch.zeros(self.numel, dtype=comm_dtype, device=device)
         self.grad_dtype = dtype
         self.comm_dtype = comm_dtype
@@ -55,11 +67,6 @@ class _Bucket:
         self.flat_grad: Optional[torch.Tensor] = (
             None if comm_dtype == dtype
             else torch.zeros(self.numel, dtype=dtype, device=device))
-        self.offsets: Dict[nn.Parameter, int] = {}
-        off = 0
-        for p in params:
-            self.offsets[p] = off
-            off += p.numel()
         self.ready = set()
         self.work: Optional[Work] = None
         self.reduced = False

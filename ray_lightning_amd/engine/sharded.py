@@ -233,14 +233,15 @@ class ShardedDDP(nn.Module):
                 p.register_post_accumulate_grad_hook(self._grad_ready))
 
     def _seal(self, owner: int, params: List[nn.Parameter]) -> None:
-        numel = sum(p.numel() for p in params)
-        flat = torch.zeros(numel, dtype=params[0].dtype,
-                           device=params[0].device)
+        from .ddp import _align8
         offsets = {}
         off = 0
         for p in params:
             offsets[p] = off
-            off += p.numel()
+            off += _align8(p.numel())
+        numel = off
+        flat = torch.zeros(numel, dtype=params[0].dtype,
+                           device=params[0].device)
         bucket = {"owner": owner, "params": params, "flat": flat,
                   "offsets": offsets, "ready": set(), "work": None,
                   "reduced": False}

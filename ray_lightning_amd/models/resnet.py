@@ -1,0 +1,101 @@
+"""ResNet-50 (bottleneck v1) — the benchmark flagship model
+(BASELINE.json configs 2-3: samples/sec, synthetic ImageNet).
+
+Own implementation (no torchvision in this environment); conv/BN/GEMM
+kernels go through MIOpen/hipBLASLt via PyTorch-ROCm —
+the framework-owned hot path is the gradient engine + fused optimizer.
+"""
+from __future__ import annotations
+
+from typing import List, Optional
+
+import torch
+import torch.nn as nn
+
+
+class Bottleneck(nn.Module):
+    expansion = 4
+
+    def __init__(self, in_ch: int, width: int, stride: int = 1,
+                 downsample: Optional[nn.Module] = None):
+        super().__init__()
+        out_ch = width * self.expansion
+        self.conv1 = nn.Conv2d(in_ch, width, 1, bias=False)
+        self.bn1 = nn.BatchNorm2d(width)
+        self.conv2 = nn.Conv2d(width, width, 3, stride=stride, padding=1,
+                               bias=False)
+        self.bn2 = nn.BatchNorm2d(width)
+        self.conv3 = nn.Conv2d(width, out_ch, 1, bias=False)
+        self.bn3 = nn.BatchNorm2d(out_ch)
+        self.relu = nn.ReLU(inplace=True)
+        self.downsample = downsample
+        self.stride = stride
+
+    def forward(self, x):
+        identity = x
+        out = self.relu(self.bn1(self.conv1(x)))
+        out = self.relu(self.bn2(self.conv2(out)))
+        out = self.bn3(self.conv3(out))
+        if self.downsample is not None:
+            identity = self.downsample(x)
+        out += identity
+        return self.relu(out)
+
+
+class ResNet(nn.Module):
+    def __init__(self, layers: List[int], num_classes: int = 1000,
+                 zero_init_residual: bool = True):
+        super().__init__()
+        self.in_ch = 64
+        self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
+        self.bn1 = nn.BatchNorm2d(64)
+        self.relu = nn.ReLU(inplace=True)
+        self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
+        self.layer1 = self._make_layer(64, layers[0])
+        self.layer2 = self._make_layer(128, layers[1], stride=2)
+        self.layer3 = self._make_layer(256, layers[2], stride=2)
+        self.layer4 = self._make_layer(512, layers[3], stride=2)
+        self.avgpool = nn.AdaptiveAvgPool2d(1)
+        self.fc = nn.Linear(512 * Bottleneck.expansion, num_classes)
+
+        for m in self.modules():
+            if isinstance(m, nn.Conv2d):
+                nn.init.kaiming_normal_(m.weight, mode="fan_out",
+                                        nonlinearity="relu")
+            elif isinstance(m, nn.BatchNorm2d):
+                nn.init.ones_(m.weight)
+                nn.init.zeros_(m.bias)
+        if zero_init_residual:
+            for m in self.modules():
+                if isinstance(m, Bottleneck):
+                    nn.init.zeros_(m.bn3.weight)
+
+    def _make_layer(self, width: int, blocks: int,
+                    stride: int = 1) -> nn.Sequential:
+        downsample = None
+        out_ch = width * Bottleneck.expansion
+        if stride != 1 or self.in_ch != out_ch:
+            downsample = nn.Sequential(
+                nn.Conv2d(self.in_ch, out_ch, 1, stride=stride,
+                          bias=False),
+                nn.BatchNorm2d(out_ch))
+        layers = [Bottleneck(self.in_ch, width, stride, downsample)]
+        self.in_ch = out_ch
+        for _ in range(1, blocks):
+            layers.append(Bottleneck(out_ch, width))
+        return nn.Sequential(*layers)
+
+    def forward(self, x):
+        x = self.maxpool(self.relu(self.bn1(self.conv1(x))))
+        x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
+        x = self.avgpool(x).flatten(1)
+        return self.fc(x)
+
+
+def resnet50(num_classes: int = 1000) -> ResNet:
+    return ResNet([3, 4, 6, 3], num_classes)
+
+
+def resnet18_like(num_classes: int = 1000) -> ResNet:
+    """Small bottleneck variant for quick tests."""
+    return ResNet([1, 1, 1, 1], num_classes)
