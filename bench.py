@@ -23,6 +23,10 @@ import os
 import sys
 import time
 
+# MIOpen exhaustive find takes minutes per conv config on a cache-cold
+# box; immediate mode costs a few % at most and is deterministic.
+os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
+
 import torch
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
@@ -78,7 +82,8 @@ def main() -> None:
         batch = min(args.batch_size, 8)
 
     torch.manual_seed(1234 + rank)
-    torch.backends.cudnn.benchmark = True
+    # benchmark=True would force MIOpen full Find (minutes, cache-cold)
+    torch.backends.cudnn.benchmark = False
 
     control = data_comm = None
     if world > 1:
