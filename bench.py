@@ -65,6 +65,9 @@ def main() -> None:
     p.add_argument("--compression", choices=["none", "bf16"],
                    default="none", help="gradient comm dtype")
     p.add_argument("--num-classes", type=int, default=1000)
+    p.add_argument("--memory-format", choices=["channels_last", "nchw"],
+                   default="channels_last",
+                   help="NHWC is the native MIOpen/CDNA4 conv layout")
     args = p.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -90,6 +93,9 @@ def main() -> None:
         control, data_comm = build_comm(rank, world, device)
 
     model = resnet50(args.num_classes).to(device)
+    channels_last = args.memory_format == "channels_last" and on_gpu
+    if channels_last:
+        model = model.to(memory_format=torch.channels_last)
     model.train()
     if world > 1:
         comm_dtype = torch.bfloat16 if args.compression == "bf16" else None
@@ -104,6 +110,8 @@ def main() -> None:
     # pre-staged synthetic batches (rotate to defeat caching)
     n_buf = 4
     images = [torch.randn(batch, 3, 224, 224, device=device)
+              .to(memory_format=torch.channels_last) if channels_last
+              else torch.randn(batch, 3, 224, 224, device=device)
               for _ in range(n_buf)]
     labels = [torch.randint(0, args.num_classes, (batch,), device=device)
               for _ in range(n_buf)]
@@ -168,6 +176,7 @@ def main() -> None:
                 "parallelism": f"dp{world}",
                 "grad_comm_dtype": args.compression,
                 "bucket_cap_mb": args.bucket_mb,
+                "memory_format": args.memory_format,
                 "device": "cuda" if on_gpu else "cpu-debug",
             },
         }
