@@ -200,6 +200,7 @@ class LightningModule(nn.Module):
         hparams = ckpt.get("hyper_parameters", {})
         hparams.update(init_kwargs)
         model = cls(**hparams)
+        model.on_load_checkpoint(ckpt)
         model.load_state_dict(ckpt["state_dict"])
         return model
 

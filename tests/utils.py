@@ -96,9 +96,12 @@ class SyntheticMNISTDataset(Dataset):
     a classifier can actually reach accuracy >= 0.5 after one epoch."""
 
     def __init__(self, num_samples: int = 512, seed: int = 1234):
+        # class centers are FIXED across splits (train/val/test must share
+        # the task); only sampling varies with `seed`.
+        gc = torch.Generator().manual_seed(999)
+        centers = torch.randn(10, 28 * 28, generator=gc) * 2.0
         g = torch.Generator().manual_seed(seed)
         self.targets = torch.randint(0, 10, (num_samples,), generator=g)
-        centers = torch.randn(10, 28 * 28, generator=g) * 2.0
         noise = torch.randn(num_samples, 28 * 28, generator=g) * 0.3
         self.data = centers[self.targets] + noise
 
