@@ -23,9 +23,11 @@ import os
 import sys
 import time
 
-# MIOpen exhaustive find takes minutes per conv config on a cache-cold
-# box; immediate mode costs a few % at most and is deterministic.
-os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
+# MIOpen find mode: the default (DYNAMIC_HYBRID) benchmarks dynamic
+# igemm solvers with real workspace during the *untimed* warmup steps.
+# FAST (immediate heuristics) picks catastrophic no-workspace wrw-conv
+# fallback kernels on MI355X (profiles/r01_resnet50_1gpu_fastfind.md).
+# Override via env if needed.
 
 import torch
 

@@ -221,6 +221,8 @@ class RayLauncher:
                 to_gpu=self._strategy.use_gpu and torch.cuda.is_available())
             trainer.model.load_state_dict(state["state_dict"])
         trainer.state = ray_output.trainer_state
+        trainer._current_epoch = ray_output.current_epoch
+        trainer._global_step = ray_output.global_step
         trainer.callback_metrics = {
             k: torch.tensor(v) for k, v in
             ray_output.callback_metrics.items()}
@@ -297,7 +299,8 @@ def _collect_rank_zero_results(trainer, results) -> Optional[_RayOutput]:
         for k, v in trainer.logged_metrics.items()}
     results = _move_to_cpu(results)
     return _RayOutput(best_model_path, state_stream, trainer.state,
-                      results, callback_metrics, logged_metrics)
+                      results, callback_metrics, logged_metrics,
+                      trainer.current_epoch, trainer.global_step)
 
 
 def _move_to_cpu(obj):

@@ -30,3 +30,8 @@ class _RayOutput(NamedTuple):
     trainer_results: Any
     callback_metrics: Dict[str, Any]
     logged_metrics: Dict[str, Any]
+    # loop counters replayed onto the driver trainer (extension over the
+    # reference's field set: our Trainer owns its loop, so the driver's
+    # epoch/step must track the worker's for repeated fit / resume).
+    current_epoch: int = 0
+    global_step: int = 0

@@ -10,11 +10,17 @@ import multiprocessing as mp
 import queue as _pyqueue
 from typing import Any, Optional
 
+import cloudpickle
+
 _MP = mp.get_context("spawn")
 
 
 class Queue:
-    """Picklable cross-process queue with non-raising ``get_nowait``."""
+    """Picklable cross-process queue with non-raising ``get_nowait``.
+
+    Items are cloudpickle-encoded: the channel carries *lambdas* (the
+    queue-of-callables worker->driver RPC, reference session.py:17-24)
+    which the manager proxy's stdlib pickle cannot serialize."""
 
     def __init__(self, _proxy=None):
         if _proxy is None:
@@ -25,14 +31,14 @@ class Queue:
             self._q = _proxy
 
     def put(self, item: Any) -> None:
-        self._q.put(item)
+        self._q.put(cloudpickle.dumps(item))
 
     def get(self, timeout: Optional[float] = None) -> Any:
-        return self._q.get(timeout=timeout)
+        return cloudpickle.loads(self._q.get(timeout=timeout))
 
     def get_nowait(self) -> Optional[Any]:
         try:
-            return self._q.get_nowait()
+            return cloudpickle.loads(self._q.get_nowait())
         except _pyqueue.Empty:
             return None
         except (EOFError, BrokenPipeError, ConnectionResetError):

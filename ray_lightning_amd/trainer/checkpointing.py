@@ -41,11 +41,16 @@ class CheckpointConnector:
             cb.on_save_checkpoint(trainer, model, checkpoint)
         return checkpoint
 
-    def save(self, filepath: str) -> None:
-        checkpoint = self.dump_checkpoint()
+    def write(self, checkpoint: Dict[str, Any], filepath: str) -> None:
         os.makedirs(os.path.dirname(os.path.abspath(filepath)),
                     exist_ok=True)
         torch.save(checkpoint, filepath)
+
+    def save(self, filepath: str) -> None:
+        """Single-process save. In distributed runs use
+        Trainer.save_checkpoint: dump_checkpoint contains collectives
+        (sharded-optimizer consolidation) so every rank must dump."""
+        self.write(self.dump_checkpoint(), filepath)
 
     def restore(self, ckpt: Union[str, Dict[str, Any]],
                 restore_training_state: bool = True) -> None:

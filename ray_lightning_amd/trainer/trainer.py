@@ -741,8 +741,12 @@ class Trainer:
     # checkpoint API
     # ------------------------------------------------------------------ #
     def save_checkpoint(self, filepath: str) -> None:
+        # dump_checkpoint can contain collectives (sharded-optimizer
+        # consolidation gathers every rank's shard), so EVERY rank must
+        # dump; only rank 0 writes.
+        checkpoint = self._checkpoint_connector.dump_checkpoint()
         if self.global_rank == 0:
-            self._checkpoint_connector.save(filepath)
+            self._checkpoint_connector.write(checkpoint, filepath)
         self.strategy.barrier()
 
     # ------------------------------------------------------------------ #

@@ -447,10 +447,14 @@ class _TuneCheckpointCallback(TuneCallback):
                 f.write(checkpoint_bytes)
 
     def _handle(self, trainer, pl_module):
-        if trainer.sanity_checking or get_actor_rank() != 0:
+        if trainer.sanity_checking:
             return
+        # dump on EVERY rank: sharded-optimizer consolidation inside
+        # dump_checkpoint is collective; only rank 0 ships the bytes.
         checkpoint_bytes = to_state_stream(
             trainer._checkpoint_connector.dump_checkpoint())
+        if get_actor_rank() != 0:
+            return
         global_step = trainer.global_step
         put_queue(lambda: self._create_checkpoint(
             checkpoint_bytes, global_step, self._filename))
