@@ -1,19 +1,23 @@
 #!/usr/bin/env python3
-"""Flagship benchmark: ResNet-50 data-parallel training throughput.
+"""Flagship benchmark: data-parallel training throughput on MI355X.
 
-BASELINE.json metric: samples/sec (whole node), ResNet-50, RayStrategy
-DDP at 1/2/4/8 MI355X workers, synthetic ImageNet-shaped data,
-random-init weights, bf16 autocast compute.
+BASELINE.json metric (default): samples/sec (whole node), ResNet-50,
+RayStrategy DDP at 1/2/4/8 MI355X workers, synthetic ImageNet-shaped
+data, random-init weights, bf16 autocast compute.
+
+Also runs BASELINE config 4 via ``--model gpt2-xl --strategy sharded``:
+GPT-2-XL bf16 weights, ShardedOptimizer (OSS 1/N optimizer state) +
+ShardedDDP reduce-to-owner gradients + sharded fused-Adam HIP kernel.
 
 Launch shapes:
   python bench.py --gpus 1 --steps K --warmup W            (single rank)
   python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
       --master-addr 127.0.0.1 bench.py --gpus N ...        (one rank/GPU)
 
-The timed region is the full training step: forward (bf16 autocast),
-backward with the NativeDDP bucketed RCCL all-reduce overlapped, bucket
-finalize + 1/world scale, fused-SGD optimizer step. Data is pre-staged
-synthetic batches (rotated) as declared in the output's "data" field.
+The timed region is the full training step: forward (bf16 autocast or
+bf16 weights), backward with the bucketed RCCL collective overlapped,
+bucket finalize + 1/world scale, fused optimizer step. Data is
+pre-staged synthetic batches (rotated) as declared in "data".
 """
 from __future__ import annotations
 
@@ -26,8 +30,8 @@ import time
 # MIOpen find mode: the default (DYNAMIC_HYBRID) benchmarks dynamic
 # igemm solvers with real workspace during the *untimed* warmup steps.
 # FAST (immediate heuristics) picks catastrophic no-workspace wrw-conv
-# fallback kernels on MI355X (profiles/r01_resnet50_1gpu_fastfind.md).
-# Override via env if needed.
+# fallback kernels on MI355X: 212 vs 6014 samples/s measured
+# (profiles/r01_resnet50_1gpu_fastfind.md). Override via env if needed.
 
 import torch
 
@@ -36,8 +40,10 @@ sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 from ray_lightning_amd.engine.comm import (TorchDistCommunicator,
                                            init_control_plane)
 from ray_lightning_amd.engine.ddp import NativeDDP
+from ray_lightning_amd.engine.sharded import ShardedDDP, ShardedOptimizer
+from ray_lightning_amd.models.gpt2 import GPT2, GPT2Config
 from ray_lightning_amd.models.resnet import resnet50
-from ray_lightning_amd.optim import FusedSGD
+from ray_lightning_amd.optim import FusedSGD, ShardedFusedAdam
 
 
 def build_comm(rank: int, world: int, device: torch.device):
@@ -61,16 +67,28 @@ def main() -> None:
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=30)
     p.add_argument("--warmup", type=int, default=10)
-    p.add_argument("--batch-size", type=int, default=256,
-                   help="per-GPU batch (weak scaling)")
+    p.add_argument("--model", choices=["resnet50", "gpt2", "gpt2-xl"],
+                   default="resnet50")
+    p.add_argument("--strategy", choices=["ddp", "sharded"], default=None,
+                   help="default: ddp for resnet50, sharded for gpt2*")
+    p.add_argument("--batch-size", type=int, default=None,
+                   help="per-GPU batch (weak scaling); default 256 "
+                        "resnet / 8 gpt2")
+    p.add_argument("--seq-len", type=int, default=1024)
     p.add_argument("--bucket-mb", type=float, default=50.0)
     p.add_argument("--compression", choices=["none", "bf16"],
-                   default="none", help="gradient comm dtype")
+                   default="none", help="gradient comm dtype (ddp)")
     p.add_argument("--num-classes", type=int, default=1000)
     p.add_argument("--memory-format", choices=["channels_last", "nchw"],
                    default="channels_last",
                    help="NHWC is the native MIOpen/CDNA4 conv layout")
     args = p.parse_args()
+
+    is_gpt = args.model.startswith("gpt2")
+    if args.strategy is None:
+        args.strategy = "sharded" if is_gpt else "ddp"
+    if args.batch_size is None:
+        args.batch_size = 8 if is_gpt else 256
 
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -87,45 +105,94 @@ def main() -> None:
         batch = min(args.batch_size, 8)
 
     torch.manual_seed(1234 + rank)
-    # benchmark=True would force MIOpen full Find (minutes, cache-cold)
+    # benchmark=True would force MIOpen exhaustive Find (minutes on a
+    # cache-cold box); default find mode is already near-optimal.
     torch.backends.cudnn.benchmark = False
 
     control = data_comm = None
     if world > 1:
         control, data_comm = build_comm(rank, world, device)
 
-    model = resnet50(args.num_classes).to(device)
-    channels_last = args.memory_format == "channels_last" and on_gpu
-    if channels_last:
-        model = model.to(memory_format=torch.channels_last)
-    model.train()
-    if world > 1:
-        comm_dtype = torch.bfloat16 if args.compression == "bf16" else None
-        wrapped = NativeDDP(model, data_comm,
-                            bucket_cap_mb=args.bucket_mb,
-                            comm_dtype=comm_dtype)
+    # ----------------------------------------------------------------- #
+    # model + optimizer + synthetic data
+    # ----------------------------------------------------------------- #
+    if is_gpt:
+        cfg = (GPT2Config.gpt2_xl() if args.model == "gpt2-xl"
+               else GPT2Config.gpt2())
+        cfg.n_positions = max(args.seq_len, 1024)
+        model = GPT2(cfg).to(device)
+        model = model.to(torch.bfloat16)
+        model.train()
+        decay = [pm for pm in model.parameters() if pm.dim() >= 2]
+        nodecay = [pm for pm in model.parameters() if pm.dim() < 2]
+        opt = ShardedFusedAdam(
+            [{"params": decay, "weight_decay": 0.1},
+             {"params": nodecay, "weight_decay": 0.0}],
+            lr=6e-4, betas=(0.9, 0.95))
+        if world > 1:
+            if args.strategy == "sharded":
+                opt = ShardedOptimizer(opt, data_comm,
+                                       bucket_cap_mb=args.bucket_mb)
+                wrapped = ShardedDDP(model, data_comm, opt,
+                                     bucket_cap_mb=args.bucket_mb)
+            else:
+                wrapped = NativeDDP(model, data_comm,
+                                    bucket_cap_mb=args.bucket_mb)
+        else:
+            wrapped = model
+        vocab = cfg.vocab_size
+        n_buf = 4
+        xs = [torch.randint(0, vocab, (batch, args.seq_len), device=device)
+              for _ in range(n_buf)]
+        ys = [torch.randint(0, vocab, (batch, args.seq_len), device=device)
+              for _ in range(n_buf)]
+
+        def fwd(i: int):
+            _, loss = model(xs[i % n_buf], ys[i % n_buf])
+            return loss
+        autocast_needed = False
+        channels_last = False
     else:
-        wrapped = model
-    opt = FusedSGD(model.parameters(), lr=0.1, momentum=0.9,
-                   weight_decay=1e-4)
+        model = resnet50(args.num_classes).to(device)
+        channels_last = args.memory_format == "channels_last" and on_gpu
+        if channels_last:
+            model = model.to(memory_format=torch.channels_last)
+        model.train()
+        opt = FusedSGD(model.parameters(), lr=0.1, momentum=0.9,
+                       weight_decay=1e-4)
+        if world > 1:
+            if args.strategy == "sharded":
+                opt = ShardedOptimizer(opt, data_comm,
+                                       bucket_cap_mb=args.bucket_mb)
+                wrapped = ShardedDDP(model, data_comm, opt,
+                                     bucket_cap_mb=args.bucket_mb)
+            else:
+                comm_dtype = (torch.bfloat16 if args.compression == "bf16"
+                              else None)
+                wrapped = NativeDDP(model, data_comm,
+                                    bucket_cap_mb=args.bucket_mb,
+                                    comm_dtype=comm_dtype)
+        else:
+            wrapped = model
+        n_buf = 4
+        images = [torch.randn(batch, 3, 224, 224, device=device)
+                  .to(memory_format=torch.channels_last) if channels_last
+                  else torch.randn(batch, 3, 224, 224, device=device)
+                  for _ in range(n_buf)]
+        labels = [torch.randint(0, args.num_classes, (batch,),
+                                device=device) for _ in range(n_buf)]
+        ac = (torch.autocast("cuda", dtype=torch.bfloat16) if on_gpu
+              else torch.autocast("cpu", torch.bfloat16))
 
-    # pre-staged synthetic batches (rotate to defeat caching)
-    n_buf = 4
-    images = [torch.randn(batch, 3, 224, 224, device=device)
-              .to(memory_format=torch.channels_last) if channels_last
-              else torch.randn(batch, 3, 224, 224, device=device)
-              for _ in range(n_buf)]
-    labels = [torch.randint(0, args.num_classes, (batch,), device=device)
-              for _ in range(n_buf)]
-
-    autocast = (torch.autocast("cuda", dtype=torch.bfloat16)
-                if on_gpu else torch.autocast("cpu", torch.bfloat16))
+        def fwd(i: int):
+            with ac:
+                logits = model(images[i % n_buf])
+                return torch.nn.functional.cross_entropy(
+                    logits.float(), labels[i % n_buf])
+        autocast_needed = True
 
     def step(i: int) -> None:
-        x, y = images[i % n_buf], labels[i % n_buf]
-        with autocast:
-            logits = model(x)
-            loss = torch.nn.functional.cross_entropy(logits.float(), y)
+        loss = fwd(i)
         loss.backward()
         if world > 1:
             wrapped.finalize_backward()
@@ -157,9 +224,26 @@ def main() -> None:
     if rank == 0:
         samples = world * batch * args.steps
         value = samples / elapsed
+        cfg_out = {
+            "model": args.model,
+            "global_batch": world * batch,
+            "parallelism": (f"dp{world}" if args.strategy == "ddp"
+                            else f"sharded-dp{world}"),
+            "bucket_cap_mb": args.bucket_mb,
+            "device": "cuda" if on_gpu else "cpu-debug",
+        }
+        if is_gpt:
+            cfg_out["seq_len"] = args.seq_len
+            cfg_out["tokens_per_s"] = round(value * args.seq_len, 1)
+            cfg_out["optimizer"] = "sharded-fused-adamw(bf16+fp32 master)"
+        else:
+            cfg_out["image_size"] = 224
+            cfg_out["grad_comm_dtype"] = args.compression
+            cfg_out["memory_format"] = args.memory_format
         out = {
             "metric": "samples/sec (whole node) ResNet-50 RayStrategy "
-                      "DDP at 1/2/4/8 MI355X workers",
+                      "DDP at 1/2/4/8 MI355X workers" if not is_gpt else
+                      f"samples/sec (whole node) {args.model} sharded",
             "value": round(value, 2),
             "unit": "samples/s",
             "n_gpus": world,
@@ -171,16 +255,7 @@ def main() -> None:
             "vs_baseline": None,
             "dtype": "bf16",
             "data": "synthetic",
-            "config": {
-                "model": "resnet50",
-                "global_batch": world * batch,
-                "image_size": 224,
-                "parallelism": f"dp{world}",
-                "grad_comm_dtype": args.compression,
-                "bucket_cap_mb": args.bucket_mb,
-                "memory_format": args.memory_format,
-                "device": "cuda" if on_gpu else "cpu-debug",
-            },
+            "config": cfg_out,
         }
         print(json.dumps(out), flush=True)
 
