@@ -1,9 +1,12 @@
 """ResNet-50 (bottleneck v1) — the benchmark flagship model
 (BASELINE.json configs 2-3: samples/sec, synthetic ImageNet).
 
-Own implementation (no torchvision in this environment); conv/BN/GEMM
-kernels go through MIOpen/hipBLASLt via PyTorch-ROCm —
-the framework-owned hot path is the gradient engine + fused optimizer.
+Own implementation (no torchvision in this environment). Convs go
+through MIOpen via PyTorch-ROCm; normalization +
+activation + residual are this framework's hand-written fused NHWC
+CDNA4 kernel (``ops/fused_bn``) — BN + elementwise were >50% of the
+MI355X step time when left to MIOpen/ATen
+(profiles/r01_resnet50_1gpu_fixedfind.md).
 """
 from __future__ import annotations
 
@@ -11,6 +14,8 @@ from typing import List, Optional
 
 import torch
 import torch.nn as nn
+
+from ..ops.fused_bn import FusedBatchNorm2d
 
 
 class Bottleneck(nn.Module):
@@ -21,25 +26,22 @@ class Bottleneck(nn.Module):
         super().__init__()
         out_ch = width * self.expansion
         self.conv1 = nn.Conv2d(in_ch, width, 1, bias=False)
-        self.bn1 = nn.BatchNorm2d(width)
+        self.bn1 = FusedBatchNorm2d(width, relu=True)
         self.conv2 = nn.Conv2d(width, width, 3, stride=stride, padding=1,
                                bias=False)
-        self.bn2 = nn.BatchNorm2d(width)
+        self.bn2 = FusedBatchNorm2d(width, relu=True)
         self.conv3 = nn.Conv2d(width, out_ch, 1, bias=False)
-        self.bn3 = nn.BatchNorm2d(out_ch)
-        self.relu = nn.ReLU(inplace=True)
+        self.bn3 = FusedBatchNorm2d(out_ch, relu=True)  # fused add+relu
         self.downsample = downsample
         self.stride = stride
 
     def forward(self, x):
         identity = x
-        out = self.relu(self.bn1(self.conv1(x)))
-        out = self.relu(self.bn2(self.conv2(out)))
-        out = self.bn3(self.conv3(out))
+        out = self.bn1(self.conv1(x))
+        out = self.bn2(self.conv2(out))
         if self.downsample is not None:
             identity = self.downsample(x)
-        out += identity
-        return self.relu(out)
+        return self.bn3(self.conv3(out), identity)
 
 
 class ResNet(nn.Module):
@@ -48,8 +50,7 @@ class ResNet(nn.Module):
         super().__init__()
         self.in_ch = 64
         self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
-        self.bn1 = nn.BatchNorm2d(64)
-        self.relu = nn.ReLU(inplace=True)
+        self.bn1 = FusedBatchNorm2d(64, relu=True)
         self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
         self.layer1 = self._make_layer(64, layers[0])
         self.layer2 = self._make_layer(128, layers[1], stride=2)
@@ -62,7 +63,7 @@ class ResNet(nn.Module):
             if isinstance(m, nn.Conv2d):
                 nn.init.kaiming_normal_(m.weight, mode="fan_out",
                                         nonlinearity="relu")
-            elif isinstance(m, nn.BatchNorm2d):
+            elif isinstance(m, nn.BatchNorm2d):  # FusedBatchNorm2d too
                 nn.init.ones_(m.weight)
                 nn.init.zeros_(m.bias)
         if zero_init_residual:
@@ -78,7 +79,7 @@ class ResNet(nn.Module):
             downsample = nn.Sequential(
                 nn.Conv2d(self.in_ch, out_ch, 1, stride=stride,
                           bias=False),
-                nn.BatchNorm2d(out_ch))
+                FusedBatchNorm2d(out_ch))
         layers = [Bottleneck(self.in_ch, width, stride, downsample)]
         self.in_ch = out_ch
         for _ in range(1, blocks):
@@ -86,7 +87,7 @@ class ResNet(nn.Module):
         return nn.Sequential(*layers)
 
     def forward(self, x):
-        x = self.maxpool(self.relu(self.bn1(self.conv1(x))))
+        x = self.maxpool(self.bn1(self.conv1(x)))
         x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
         x = self.avgpool(x).flatten(1)
         return self.fc(x)

@@ -29,7 +29,8 @@ def build(verbose: bool = False) -> None:
     common_cflags = ["-O3", "-std=c++17"]
 
     for name, sources, ldflags in (
-        ("_hip_ops", [os.path.join(_CSRC, "hip_ops.hip")], []),
+        ("_hip_ops", [os.path.join(_CSRC, "hip_ops.hip"),
+                      os.path.join(_CSRC, "fused_bn.hip")], []),
         ("_rccl_comm", [os.path.join(_CSRC, "rccl_comm.hip")],
          ["-L/opt/rocm/lib", "-lrccl"]),
     ):

@@ -1,0 +1,514 @@
+// Fused BatchNorm(+ReLU)(+residual add) for NHWC (channels_last) on
+// CDNA4. Replaces the MIOpen BN kernel quintet + ATen add/clamp chain
+// that dominates the ResNet-50 step (see
+// profiles/r01_resnet50_1gpu_fixedfind.md: BN + elementwise = >50% of
+// step time; the convs themselves are only ~1/3).
+//
+// Memory-bound design: the train forward is 2 passes over x (stats +
+// apply) instead of eager's 4 reads + 3 writes (BN stats, BN norm,
+// add, relu); the backward is 2 passes (reduce + dx) instead of 5.
+// All accumulation in fp32; bf16 storage with 16B (8-lane) vector
+// access; per-channel reductions staged through LDS then one fp32
+// atomicAdd per (workgroup, channel).
+//
+// Layout convention: x is [R, C] row-major with C contiguous
+// (R = N*H*W): exactly torch channels_last.
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#define CHECK_HIP_BN(cmd)                                                  \
+  do {                                                                     \
+    hipError_t e = (cmd);                                                  \
+    TORCH_CHECK(e == hipSuccess, "HIP error: ", hipGetErrorString(e));     \
+  } while (0)
+
+namespace {
+
+constexpr int kT = 256;  // 4 wave64 per workgroup
+
+template <typename T> struct BnVec;
+template <> struct BnVec<float> { static constexpr int V = 4; };
+template <> struct BnVec<__hip_bfloat16> { static constexpr int V = 8; };
+
+template <typename T>
+__device__ __forceinline__ float bn_tof(T v);
+template <>
+__device__ __forceinline__ float bn_tof<float>(float v) { return v; }
+template <>
+__device__ __forceinline__ float bn_tof<__hip_bfloat16>(
+    __hip_bfloat16 v) { return __bfloat162float(v); }
+
+template <typename T>
+__device__ __forceinline__ T bn_fromf(float v);
+template <>
+__device__ __forceinline__ float bn_fromf<float>(float v) { return v; }
+template <>
+__device__ __forceinline__ __hip_bfloat16 bn_fromf<__hip_bfloat16>(
+    float v) { return __float2bfloat16(v); }
+
+template <typename T, int V>
+__device__ __forceinline__ void load_vec(T (&dst)[V], const T* src) {
+  *reinterpret_cast<int4*>(dst) = *reinterpret_cast<const int4*>(src);
+}
+template <typename T, int V>
+__device__ __forceinline__ void store_vec(T* dst, const T (&src)[V]) {
+  *reinterpret_cast<int4*>(dst) = *reinterpret_cast<const int4*>(src);
+}
+
+// --------------------------------------------------------------------
+// stats: per-channel sum / sumsq (fp32 atomics into zeroed buffers)
+// --------------------------------------------------------------------
+template <typename T>
+__global__ __launch_bounds__(kT) void bn_stats_kernel(
+    const T* __restrict__ x, float* __restrict__ sum,
+    float* __restrict__ sumsq, long R, int C) {
+  constexpr int V = BnVec<T>::V;
+  const int tpr = C / V;             // threads per row (<= kT)
+  const int rpb = kT / tpr;          // rows per block (power of 2)
+  const int slot = threadIdx.x % tpr;
+  const int row_in_block = threadIdx.x / tpr;
+  const int c0 = slot * V;
+
+  float acc[V];
+  float acc2[V];
+  #pragma unroll
+  for (int i = 0; i < V; ++i) { acc[i] = 0.f; acc2[i] = 0.f; }
+
+  const long row_stride = (long)gridDim.x * rpb;
+  for (long r = (long)blockIdx.x * rpb + row_in_block; r < R;
+       r += row_stride) {
+    T v[V];
+    load_vec<T, V>(v, x + r * C + c0);
+    #pragma unroll
+    for (int i = 0; i < V; ++i) {
+      const float f = bn_tof<T>(v[i]);
+      acc[i] += f;
+      acc2[i] += f * f;
+    }
+  }
+
+  // LDS reduce across the rpb rows sharing each channel slot
+  __shared__ float lds[kT * BnVec<T>::V];
+  #pragma unroll
+  for (int i = 0; i < V; ++i) lds[threadIdx.x * V + i] = acc[i];
+  __syncthreads();
+  for (int s = rpb / 2; s > 0; s >>= 1) {
+    if (row_in_block < s) {
+      #pragma unroll
+      for (int i = 0; i < V; ++i)
+        lds[threadIdx.x * V + i] += lds[(threadIdx.x + s * tpr) * V + i];
+    }
+    __syncthreads();
+  }
+  if (row_in_block == 0) {
+    #pragma unroll
+    for (int i = 0; i < V; ++i)
+      atomicAdd(&sum[c0 + i], lds[threadIdx.x * V + i]);
+  }
+  __syncthreads();
+  #pragma unroll
+  for (int i = 0; i < V; ++i) lds[threadIdx.x * V + i] = acc2[i];
+  __syncthreads();
+  for (int s = rpb / 2; s > 0; s >>= 1) {
+    if (row_in_block < s) {
+      #pragma unroll
+      for (int i = 0; i < V; ++i)
+        lds[threadIdx.x * V + i] += lds[(threadIdx.x + s * tpr) * V + i];
+    }
+    __syncthreads();
+  }
+  if (row_in_block == 0) {
+    #pragma unroll
+    for (int i = 0; i < V; ++i)
+      atomicAdd(&sumsq[c0 + i], lds[threadIdx.x * V + i]);
+  }
+}
+
+// --------------------------------------------------------------------
+// finalize stats -> mean/invstd + scale/bias (+ running stats update)
+// --------------------------------------------------------------------
+__global__ void bn_finalize_kernel(
+    const float* __restrict__ sum, const float* __restrict__ sumsq,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    float* __restrict__ running_mean, float* __restrict__ running_var,
+    float* __restrict__ mean_out, float* __restrict__ invstd_out,
+    float* __restrict__ scale_out, float* __restrict__ bias_out,
+    long R, int C, float momentum, float eps) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const float m = sum[c] / (float)R;
+  const float var = fmaxf(sumsq[c] / (float)R - m * m, 0.f);
+  const float inv = rsqrtf(var + eps);
+  mean_out[c] = m;
+  invstd_out[c] = inv;
+  const float sc = gamma[c] * inv;
+  scale_out[c] = sc;
+  bias_out[c] = beta[c] - m * sc;
+  if (running_mean != nullptr) {
+    // unbiased variance for running stats (torch semantics)
+    const float var_unb = R > 1 ? var * (float)R / (float)(R - 1) : var;
+    running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * m;
+    running_var[c] = (1.f - momentum) * running_var[c] +
+                     momentum * var_unb;
+  }
+}
+
+// eval mode: scale/bias straight from running stats
+__global__ void bn_eval_coef_kernel(
+    const float* __restrict__ running_mean,
+    const float* __restrict__ running_var,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    float* __restrict__ scale_out, float* __restrict__ bias_out, int C,
+    float eps) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const float inv = rsqrtf(running_var[c] + eps);
+  const float sc = gamma[c] * inv;
+  scale_out[c] = sc;
+  bias_out[c] = beta[c] - running_mean[c] * sc;
+}
+
+// --------------------------------------------------------------------
+// apply: y = [relu](scale*x + bias [+ res])
+// --------------------------------------------------------------------
+template <typename T, bool RELU, bool RES>
+__global__ __launch_bounds__(kT) void bn_apply_kernel(
+    const T* __restrict__ x, T* __restrict__ y,
+    const float* __restrict__ scale, const float* __restrict__ bias,
+    const T* __restrict__ res, long total, int C) {
+  constexpr int V = BnVec<T>::V;
+  const long stride = (long)gridDim.x * kT;
+  const long nvec = total / V;
+  for (long vi = (long)blockIdx.x * kT + threadIdx.x; vi < nvec;
+       vi += stride) {
+    const long idx = vi * V;
+    const int c0 = (int)(idx % C);
+    T xv[V];
+    load_vec<T, V>(xv, x + idx);
+    T rv[V];
+    if constexpr (RES) load_vec<T, V>(rv, res + idx);
+    T ov[V];
+    #pragma unroll
+    for (int i = 0; i < V; ++i) {
+      float o = scale[c0 + i] * bn_tof<T>(xv[i]) + bias[c0 + i];
+      if constexpr (RES) o += bn_tof<T>(rv[i]);
+      if constexpr (RELU) o = fmaxf(o, 0.f);
+      ov[i] = bn_fromf<T>(o);
+    }
+    store_vec<T, V>(y + idx, ov);
+  }
+}
+
+// --------------------------------------------------------------------
+// backward reduce: per-channel sum(dy_eff), sum(dy_eff * xhat)
+// dy_eff = relu ? dy * (y > 0) : dy
+// --------------------------------------------------------------------
+template <typename T, bool RELU>
+__global__ __launch_bounds__(kT) void bn_bwd_reduce_kernel(
+    const T* __restrict__ dy, const T* __restrict__ y,
+    const T* __restrict__ x, const float* __restrict__ mean,
+    const float* __restrict__ invstd, float* __restrict__ dsum,
+    float* __restrict__ dxhat_sum, long R, int C) {
+  constexpr int V = BnVec<T>::V;
+  const int tpr = C / V;
+  const int rpb = kT / tpr;
+  const int slot = threadIdx.x % tpr;
+  const int row_in_block = threadIdx.x / tpr;
+  const int c0 = slot * V;
+
+  float mu[V], is[V];
+  #pragma unroll
+  for (int i = 0; i < V; ++i) {
+    mu[i] = mean[c0 + i];
+    is[i] = invstd[c0 + i];
+  }
+
+  float a0[V], a1[V];
+  #pragma unroll
+  for (int i = 0; i < V; ++i) { a0[i] = 0.f; a1[i] = 0.f; }
+
+  const long row_stride = (long)gridDim.x * rpb;
+  for (long r = (long)blockIdx.x * rpb + row_in_block; r < R;
+       r += row_stride) {
+    T dv[V], xv[V];
+    load_vec<T, V>(dv, dy + r * C + c0);
+    load_vec<T, V>(xv, x + r * C + c0);
+    T yv[V];
+    if constexpr (RELU) load_vec<T, V>(yv, y + r * C + c0);
+    #pragma unroll
+    for (int i = 0; i < V; ++i) {
+      float d = bn_tof<T>(dv[i]);
+      if constexpr (RELU) d = bn_tof<T>(yv[i]) > 0.f ? d : 0.f;
+      const float xh = (bn_tof<T>(xv[i]) - mu[i]) * is[i];
+      a0[i] += d;
+      a1[i] += d * xh;
+    }
+  }
+
+  __shared__ float lds[kT * BnVec<T>::V];
+  #pragma unroll
+  for (int i = 0; i < V; ++i) lds[threadIdx.x * V + i] = a0[i];
+  __syncthreads();
+  for (int s = rpb / 2; s > 0; s >>= 1) {
+    if (row_in_block < s) {
+      #pragma unroll
+      for (int i = 0; i < V; ++i)
+        lds[threadIdx.x * V + i] += lds[(threadIdx.x + s * tpr) * V + i];
+    }
+    __syncthreads();
+  }
+  if (row_in_block == 0) {
+    #pragma unroll
+    for (int i = 0; i < V; ++i)
+      atomicAdd(&dsum[c0 + i], lds[threadIdx.x * V + i]);
+  }
+  __syncthreads();
+  #pragma unroll
+  for (int i = 0; i < V; ++i) lds[threadIdx.x * V + i] = a1[i];
+  __syncthreads();
+  for (int s = rpb / 2; s > 0; s >>= 1) {
+    if (row_in_block < s) {
+      #pragma unroll
+      for (int i = 0; i < V; ++i)
+        lds[threadIdx.x * V + i] += lds[(threadIdx.x + s * tpr) * V + i];
+    }
+    __syncthreads();
+  }
+  if (row_in_block == 0) {
+    #pragma unroll
+    for (int i = 0; i < V; ++i)
+      atomicAdd(&dxhat_sum[c0 + i], lds[threadIdx.x * V + i]);
+  }
+}
+
+// --------------------------------------------------------------------
+// backward dx: dx = gamma*invstd * (dy_eff - dsum/R - xhat*dxhat_sum/R)
+// optional dres = dy_eff (gradient of the fused residual input)
+// --------------------------------------------------------------------
+template <typename T, bool RELU, bool RES>
+__global__ __launch_bounds__(kT) void bn_bwd_dx_kernel(
+    const T* __restrict__ dy, const T* __restrict__ y,
+    const T* __restrict__ x, const float* __restrict__ mean,
+    const float* __restrict__ invstd, const float* __restrict__ gamma,
+    const float* __restrict__ dsum, const float* __restrict__ dxhat_sum,
+    T* __restrict__ dx, T* __restrict__ dres, long total, int C,
+    float invR) {
+  constexpr int V = BnVec<T>::V;
+  const long stride = (long)gridDim.x * kT;
+  const long nvec = total / V;
+  for (long vi = (long)blockIdx.x * kT + threadIdx.x; vi < nvec;
+       vi += stride) {
+    const long idx = vi * V;
+    const int c0 = (int)(idx % C);
+    T dv[V], xv[V];
+    load_vec<T, V>(dv, dy + idx);
+    load_vec<T, V>(xv, x + idx);
+    T yv[V];
+    if constexpr (RELU) load_vec<T, V>(yv, y + idx);
+    T dxv[V];
+    T drv[V];
+    #pragma unroll
+    for (int i = 0; i < V; ++i) {
+      const int c = c0 + i;
+      float d = bn_tof<T>(dv[i]);
+      if constexpr (RELU) d = bn_tof<T>(yv[i]) > 0.f ? d : 0.f;
+      if constexpr (RES) drv[i] = bn_fromf<T>(d);
+      const float xh = (bn_tof<T>(xv[i]) - mean[c]) * invstd[c];
+      const float g = gamma[c] * invstd[c] *
+          (d - dsum[c] * invR - xh * dxhat_sum[c] * invR);
+      dxv[i] = bn_fromf<T>(g);
+    }
+    store_vec<T, V>(dx + idx, dxv);
+    if constexpr (RES) store_vec<T, V>(dres + idx, drv);
+  }
+}
+
+long bn_grid_rows(long R, int rpb) {
+  long blocks = (R + rpb - 1) / rpb;
+  return std::min<long>(std::max<long>(blocks, 1), 8192);
+}
+
+long bn_grid_elems(long nvec) {
+  long blocks = (nvec + kT - 1) / kT;
+  return std::min<long>(std::max<long>(blocks, 1), 16384);
+}
+
+template <typename T>
+void bn_fwd_impl(const torch::Tensor& x, torch::Tensor& y,
+                 const torch::Tensor& gamma, const torch::Tensor& beta,
+                 torch::Tensor& running_mean, torch::Tensor& running_var,
+                 torch::Tensor& mean, torch::Tensor& invstd,
+                 const c10::optional<torch::Tensor>& res, bool relu,
+                 bool training, double momentum, double eps, long R,
+                 int C, hipStream_t stream) {
+  auto opts = gamma.options().dtype(at::kFloat);
+  auto scale = torch::empty({C}, opts);
+  auto bias = torch::empty({C}, opts);
+  if (training) {
+    auto sum = torch::zeros({C}, opts);
+    auto sumsq = torch::zeros({C}, opts);
+    const int tpr = C / BnVec<T>::V;
+    const int rpb = kT / tpr;
+    hipLaunchKernelGGL((bn_stats_kernel<T>),
+                       dim3(bn_grid_rows(R, rpb)), dim3(kT), 0, stream,
+                       reinterpret_cast<const T*>(x.data_ptr()),
+                       sum.data_ptr<float>(), sumsq.data_ptr<float>(), R,
+                       C);
+    CHECK_HIP_BN(hipGetLastError());
+    hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256),
+                       dim3(256), 0, stream, sum.data_ptr<float>(),
+                       sumsq.data_ptr<float>(), gamma.data_ptr<float>(),
+                       beta.data_ptr<float>(),
+                       running_mean.defined()
+                           ? running_mean.data_ptr<float>() : nullptr,
+                       running_var.defined()
+                           ? running_var.data_ptr<float>() : nullptr,
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       scale.data_ptr<float>(), bias.data_ptr<float>(),
+                       R, C, (float)momentum, (float)eps);
+    CHECK_HIP_BN(hipGetLastError());
+  } else {
+    hipLaunchKernelGGL(bn_eval_coef_kernel, dim3((C + 255) / 256),
+                       dim3(256), 0, stream,
+                       running_mean.data_ptr<float>(),
+                       running_var.data_ptr<float>(),
+                       gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                       scale.data_ptr<float>(), bias.data_ptr<float>(),
+                       C, (float)eps);
+    CHECK_HIP_BN(hipGetLastError());
+  }
+  const long total = R * (long)C;
+  const long grid = bn_grid_elems(total / BnVec<T>::V);
+  const T* resp = res.has_value()
+      ? reinterpret_cast<const T*>(res->data_ptr()) : nullptr;
+  #define APPLY(RELU_, RES_)                                            \
+    hipLaunchKernelGGL((bn_apply_kernel<T, RELU_, RES_>), dim3(grid),   \
+                       dim3(kT), 0, stream,                             \
+                       reinterpret_cast<const T*>(x.data_ptr()),        \
+                       reinterpret_cast<T*>(y.data_ptr()),              \
+                       scale.data_ptr<float>(), bias.data_ptr<float>(), \
+                       resp, total, C)
+  if (relu && resp) APPLY(true, true);
+  else if (relu) APPLY(true, false);
+  else if (resp) APPLY(false, true);
+  else APPLY(false, false);
+  #undef APPLY
+  CHECK_HIP_BN(hipGetLastError());
+}
+
+template <typename T>
+void bn_bwd_impl(const torch::Tensor& dy, const torch::Tensor& y,
+                 const torch::Tensor& x, const torch::Tensor& mean,
+                 const torch::Tensor& invstd, const torch::Tensor& gamma,
+                 torch::Tensor& dx, torch::Tensor& dgamma,
+                 torch::Tensor& dbeta,
+                 const c10::optional<torch::Tensor>& dres, bool relu,
+                 long R, int C, hipStream_t stream) {
+  const int tpr = C / BnVec<T>::V;
+  const int rpb = kT / tpr;
+  // dgamma/dbeta double as the reduce accumulators (zeroed by caller)
+  #define RED(RELU_)                                                     \
+    hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, RELU_>),                 \
+                       dim3(bn_grid_rows(R, rpb)), dim3(kT), 0, stream,  \
+                       reinterpret_cast<const T*>(dy.data_ptr()),        \
+                       reinterpret_cast<const T*>(y.data_ptr()),         \
+                       reinterpret_cast<const T*>(x.data_ptr()),         \
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(), \
+                       dbeta.data_ptr<float>(),                          \
+                       dgamma.data_ptr<float>(), R, C)
+  if (relu) RED(true); else RED(false);
+  #undef RED
+  CHECK_HIP_BN(hipGetLastError());
+
+  const long total = R * (long)C;
+  const long grid = bn_grid_elems(total / BnVec<T>::V);
+  T* dresp = dres.has_value()
+      ? reinterpret_cast<T*>(dres->data_ptr()) : nullptr;
+  const float invR = 1.f / (float)R;
+  #define DX(RELU_, RES_)                                                \
+    hipLaunchKernelGGL((bn_bwd_dx_kernel<T, RELU_, RES_>), dim3(grid),   \
+                       dim3(kT), 0, stream,                              \
+                       reinterpret_cast<const T*>(dy.data_ptr()),        \
+                       reinterpret_cast<const T*>(y.data_ptr()),         \
+                       reinterpret_cast<const T*>(x.data_ptr()),         \
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(), \
+                       gamma.data_ptr<float>(),                          \
+                       dbeta.data_ptr<float>(),                          \
+                       dgamma.data_ptr<float>(), \
+                       reinterpret_cast<T*>(dx.data_ptr()), dresp,       \
+                       total, C, invR)
+  if (relu && dresp) DX(true, true);
+  else if (relu) DX(true, false);
+  else if (dresp) DX(false, true);
+  else DX(false, false);
+  #undef DX
+  CHECK_HIP_BN(hipGetLastError());
+}
+
+}  // namespace
+
+// x: [N, C, H, W] channels_last (viewed as [R, C]); gamma/beta fp32.
+// Returns (y, save_mean, save_invstd).
+std::vector<torch::Tensor> fused_bn_fwd(
+    torch::Tensor x, torch::Tensor gamma, torch::Tensor beta,
+    torch::Tensor running_mean, torch::Tensor running_var,
+    c10::optional<torch::Tensor> res, bool relu, bool training,
+    double momentum, double eps) {
+  TORCH_CHECK(x.dim() == 4, "fused_bn: 4-D input expected");
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "fused_bn: channels_last input required");
+  const int C = (int)x.size(1);
+  const long R = x.numel() / C;
+  const int V = x.scalar_type() == at::kBFloat16 ? 8 : 4;
+  TORCH_CHECK(C % V == 0 && C / V <= kT,
+              "fused_bn: C=", C, " unsupported for vec", V);
+  if (res.has_value()) {
+    TORCH_CHECK(res->is_contiguous(at::MemoryFormat::ChannelsLast) &&
+                res->scalar_type() == x.scalar_type() &&
+                res->sizes() == x.sizes(), "fused_bn: bad residual");
+  }
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  auto y = torch::empty_like(x);  // keeps channels_last layout
+  auto fopts = gamma.options().dtype(at::kFloat);
+  auto mean = torch::empty({C}, fopts);
+  auto invstd = torch::empty({C}, fopts);
+  if (x.scalar_type() == at::kBFloat16)
+    bn_fwd_impl<__hip_bfloat16>(x, y, gamma, beta, running_mean,
+                                running_var, mean, invstd, res, relu,
+                                training, momentum, eps, R, C, stream);
+  else if (x.scalar_type() == at::kFloat)
+    bn_fwd_impl<float>(x, y, gamma, beta, running_mean, running_var,
+                       mean, invstd, res, relu, training, momentum, eps,
+                       R, C, stream);
+  else
+    TORCH_CHECK(false, "fused_bn: dtype must be bf16 or f32");
+  return {y, mean, invstd};
+}
+
+// Returns (dx, dgamma, dbeta[, dres]).
+std::vector<torch::Tensor> fused_bn_bwd(
+    torch::Tensor dy, torch::Tensor y, torch::Tensor x,
+    torch::Tensor mean, torch::Tensor invstd, torch::Tensor gamma,
+    bool relu, bool has_res) {
+  const int C = (int)x.size(1);
+  const long R = x.numel() / C;
+  dy = dy.contiguous(at::MemoryFormat::ChannelsLast);
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  auto dx = torch::empty_like(x);
+  auto fopts = gamma.options().dtype(at::kFloat);
+  auto dgamma = torch::zeros({C}, fopts);
+  auto dbeta = torch::zeros({C}, fopts);
+  c10::optional<torch::Tensor> dres;
+  if (has_res) dres = torch::empty_like(x);
+  if (x.scalar_type() == at::kBFloat16)
+    bn_bwd_impl<__hip_bfloat16>(dy, y, x, mean, invstd, gamma, dx,
+                                dgamma, dbeta, dres, relu, R, C, stream);
+  else
+    bn_bwd_impl<float>(dy, y, x, mean, invstd, gamma, dx, dgamma, dbeta,
+                       dres, relu, R, C, stream);
+  std::vector<torch::Tensor> out = {dx, dgamma, dbeta};
+  if (has_res) out.push_back(*dres);
+  return out;
+}

@@ -588,9 +588,24 @@ void sharded_adam(std::vector<torch::Tensor> params_bf16,
   }
 }
 
+// fused BN(+ReLU)(+residual) for channels_last — defined in fused_bn.hip
+std::vector<torch::Tensor> fused_bn_fwd(
+    torch::Tensor x, torch::Tensor gamma, torch::Tensor beta,
+    torch::Tensor running_mean, torch::Tensor running_var,
+    c10::optional<torch::Tensor> res, bool relu, bool training,
+    double momentum, double eps);
+std::vector<torch::Tensor> fused_bn_bwd(
+    torch::Tensor dy, torch::Tensor y, torch::Tensor x,
+    torch::Tensor mean, torch::Tensor invstd, torch::Tensor gamma,
+    bool relu, bool has_res);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("multi_tensor_pack", &multi_tensor_pack,
         "multi-tensor flatten/cast into bucket slots");
+  m.def("fused_bn_fwd", &fused_bn_fwd,
+        "fused NHWC BN(+ReLU)(+residual) forward");
+  m.def("fused_bn_bwd", &fused_bn_bwd,
+        "fused NHWC BN(+ReLU)(+residual) backward");
   m.def("scale_inplace", &scale_inplace, "flat *= s");
   m.def("scale_cast", &scale_cast, "dst_f32 = src_bf16 * s");
   m.def("fused_sgd", &fused_sgd, "fused multi-tensor SGD(momentum)");
