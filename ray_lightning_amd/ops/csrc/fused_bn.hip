@@ -7,9 +7,15 @@
 // Memory-bound design: the train forward is 2 passes over x (stats +
 // apply) instead of eager's 4 reads + 3 writes (BN stats, BN norm,
 // add, relu); the backward is 2 passes (reduce + dx) instead of 5.
-// All accumulation in fp32; bf16 storage with 16B (8-lane) vector
-// access; per-channel reductions staged through LDS then one fp32
-// atomicAdd per (workgroup, channel).
+//
+// Per-channel reductions are TWO-STAGE: stage 1 writes per-workgroup
+// partials [nblocks, C] (coalesced stores, no atomics — a v1 of this
+// file used one fp32 atomicAdd per (block, channel) and the ~8k
+// serialized RMWs per address made stats 50x slower than MIOpen's);
+// stage 2 folds partials and emits the per-channel coefficients the
+// elementwise pass needs, so the hot elementwise kernels do float4
+// coefficient loads instead of 16 scalar loads per 8 elements.
+// Fixed stage-1 grid => deterministic accumulation order.
 //
 // Layout convention: x is [R, C] row-major with C contiguous
 // (R = N*H*W): exactly torch channels_last.
@@ -26,7 +32,8 @@
 
 namespace {
 
-constexpr int kT = 256;  // 4 wave64 per workgroup
+constexpr int kT = 256;        // 4 wave64 per workgroup
+constexpr int kMaxStage1 = 1024;  // stage-1 workgroups (reduction grid)
 
 template <typename T> struct BnVec;
 template <> struct BnVec<float> { static constexpr int V = 4; };
@@ -56,14 +63,23 @@ template <typename T, int V>
 __device__ __forceinline__ void store_vec(T* dst, const T (&src)[V]) {
   *reinterpret_cast<int4*>(dst) = *reinterpret_cast<const int4*>(src);
 }
+template <int V>
+__device__ __forceinline__ void load_coef(float (&dst)[V],
+                                          const float* src) {
+  #pragma unroll
+  for (int i = 0; i < V; i += 4)
+    *reinterpret_cast<float4*>(dst + i) =
+        *reinterpret_cast<const float4*>(src + i);
+}
 
 // --------------------------------------------------------------------
-// stats: per-channel sum / sumsq (fp32 atomics into zeroed buffers)
+// stage 1 stats: per-(block, channel) partial sum / sumsq
+// partial layout: [2, nblocks, C] (sum plane then sumsq plane)
 // --------------------------------------------------------------------
 template <typename T>
 __global__ __launch_bounds__(kT) void bn_stats_kernel(
-    const T* __restrict__ x, float* __restrict__ sum,
-    float* __restrict__ sumsq, long R, int C) {
+    const T* __restrict__ x, float* __restrict__ partial, long R,
+    int C) {
   constexpr int V = BnVec<T>::V;
   const int tpr = C / V;             // threads per row (<= kT)
   const int rpb = kT / tpr;          // rows per block (power of 2)
@@ -71,8 +87,7 @@ __global__ __launch_bounds__(kT) void bn_stats_kernel(
   const int row_in_block = threadIdx.x / tpr;
   const int c0 = slot * V;
 
-  float acc[V];
-  float acc2[V];
+  float acc[V], acc2[V];
   #pragma unroll
   for (int i = 0; i < V; ++i) { acc[i] = 0.f; acc2[i] = 0.f; }
 
@@ -89,48 +104,38 @@ __global__ __launch_bounds__(kT) void bn_stats_kernel(
     }
   }
 
-  // LDS reduce across the rpb rows sharing each channel slot
   __shared__ float lds[kT * BnVec<T>::V];
-  #pragma unroll
-  for (int i = 0; i < V; ++i) lds[threadIdx.x * V + i] = acc[i];
-  __syncthreads();
-  for (int s = rpb / 2; s > 0; s >>= 1) {
-    if (row_in_block < s) {
-      #pragma unroll
-      for (int i = 0; i < V; ++i)
-        lds[threadIdx.x * V + i] += lds[(threadIdx.x + s * tpr) * V + i];
-    }
-    __syncthreads();
-  }
-  if (row_in_block == 0) {
+  const long nbC = (long)gridDim.x * C;
+  // reduce sum then sumsq across the rpb rows sharing a channel slot
+  for (int pass = 0; pass < 2; ++pass) {
     #pragma unroll
     for (int i = 0; i < V; ++i)
-      atomicAdd(&sum[c0 + i], lds[threadIdx.x * V + i]);
-  }
-  __syncthreads();
-  #pragma unroll
-  for (int i = 0; i < V; ++i) lds[threadIdx.x * V + i] = acc2[i];
-  __syncthreads();
-  for (int s = rpb / 2; s > 0; s >>= 1) {
-    if (row_in_block < s) {
+      lds[threadIdx.x * V + i] = pass ? acc2[i] : acc[i];
+    __syncthreads();
+    for (int s = rpb / 2; s > 0; s >>= 1) {
+      if (row_in_block < s) {
+        #pragma unroll
+        for (int i = 0; i < V; ++i)
+          lds[threadIdx.x * V + i] +=
+              lds[(threadIdx.x + s * tpr) * V + i];
+      }
+      __syncthreads();
+    }
+    if (row_in_block == 0) {
       #pragma unroll
       for (int i = 0; i < V; ++i)
-        lds[threadIdx.x * V + i] += lds[(threadIdx.x + s * tpr) * V + i];
+        partial[pass * nbC + (long)blockIdx.x * C + c0 + i] =
+            lds[threadIdx.x * V + i];
     }
     __syncthreads();
-  }
-  if (row_in_block == 0) {
-    #pragma unroll
-    for (int i = 0; i < V; ++i)
-      atomicAdd(&sumsq[c0 + i], lds[threadIdx.x * V + i]);
   }
 }
 
 // --------------------------------------------------------------------
-// finalize stats -> mean/invstd + scale/bias (+ running stats update)
+// stage 2: fold partials -> mean/invstd + scale/bias (+ running)
 // --------------------------------------------------------------------
-__global__ void bn_finalize_kernel(
-    const float* __restrict__ sum, const float* __restrict__ sumsq,
+__global__ void bn_stats_fold_kernel(
+    const float* __restrict__ partial, int nblocks,
     const float* __restrict__ gamma, const float* __restrict__ beta,
     float* __restrict__ running_mean, float* __restrict__ running_var,
     float* __restrict__ mean_out, float* __restrict__ invstd_out,
@@ -138,8 +143,14 @@ __global__ void bn_finalize_kernel(
     long R, int C, float momentum, float eps) {
   const int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  const float m = sum[c] / (float)R;
-  const float var = fmaxf(sumsq[c] / (float)R - m * m, 0.f);
+  const long nbC = (long)nblocks * C;
+  float s = 0.f, s2 = 0.f;
+  for (int b = 0; b < nblocks; ++b) {
+    s += partial[(long)b * C + c];
+    s2 += partial[nbC + (long)b * C + c];
+  }
+  const float m = s / (float)R;
+  const float var = fmaxf(s2 / (float)R - m * m, 0.f);
   const float inv = rsqrtf(var + eps);
   mean_out[c] = m;
   invstd_out[c] = inv;
@@ -187,12 +198,15 @@ __global__ __launch_bounds__(kT) void bn_apply_kernel(
     const int c0 = (int)(idx % C);
     T xv[V];
     load_vec<T, V>(xv, x + idx);
+    float sc[V], bi[V];
+    load_coef<V>(sc, scale + c0);
+    load_coef<V>(bi, bias + c0);
     T rv[V];
     if constexpr (RES) load_vec<T, V>(rv, res + idx);
     T ov[V];
     #pragma unroll
     for (int i = 0; i < V; ++i) {
-      float o = scale[c0 + i] * bn_tof<T>(xv[i]) + bias[c0 + i];
+      float o = sc[i] * bn_tof<T>(xv[i]) + bi[i];
       if constexpr (RES) o += bn_tof<T>(rv[i]);
       if constexpr (RELU) o = fmaxf(o, 0.f);
       ov[i] = bn_fromf<T>(o);
@@ -202,15 +216,16 @@ __global__ __launch_bounds__(kT) void bn_apply_kernel(
 }
 
 // --------------------------------------------------------------------
-// backward reduce: per-channel sum(dy_eff), sum(dy_eff * xhat)
-// dy_eff = relu ? dy * (y > 0) : dy
+// backward stage 1: per-(block, channel) partial sum(dy_eff) and
+// sum(dy_eff * xhat); dy_eff = relu ? dy * (y > 0) : dy
+// partial layout: [2, nblocks, C]
 // --------------------------------------------------------------------
 template <typename T, bool RELU>
 __global__ __launch_bounds__(kT) void bn_bwd_reduce_kernel(
     const T* __restrict__ dy, const T* __restrict__ y,
     const T* __restrict__ x, const float* __restrict__ mean,
-    const float* __restrict__ invstd, float* __restrict__ dsum,
-    float* __restrict__ dxhat_sum, long R, int C) {
+    const float* __restrict__ invstd, float* __restrict__ partial,
+    long R, int C) {
   constexpr int V = BnVec<T>::V;
   const int tpr = C / V;
   const int rpb = kT / tpr;
@@ -219,11 +234,8 @@ __global__ __launch_bounds__(kT) void bn_bwd_reduce_kernel(
   const int c0 = slot * V;
 
   float mu[V], is[V];
-  #pragma unroll
-  for (int i = 0; i < V; ++i) {
-    mu[i] = mean[c0 + i];
-    is[i] = invstd[c0 + i];
-  }
+  load_coef<V>(mu, mean + c0);
+  load_coef<V>(is, invstd + c0);
 
   float a0[V], a1[V];
   #pragma unroll
@@ -248,53 +260,73 @@ __global__ __launch_bounds__(kT) void bn_bwd_reduce_kernel(
   }
 
   __shared__ float lds[kT * BnVec<T>::V];
-  #pragma unroll
-  for (int i = 0; i < V; ++i) lds[threadIdx.x * V + i] = a0[i];
-  __syncthreads();
-  for (int s = rpb / 2; s > 0; s >>= 1) {
-    if (row_in_block < s) {
-      #pragma unroll
-      for (int i = 0; i < V; ++i)
-        lds[threadIdx.x * V + i] += lds[(threadIdx.x + s * tpr) * V + i];
-    }
-    __syncthreads();
-  }
-  if (row_in_block == 0) {
+  const long nbC = (long)gridDim.x * C;
+  for (int pass = 0; pass < 2; ++pass) {
     #pragma unroll
     for (int i = 0; i < V; ++i)
-      atomicAdd(&dsum[c0 + i], lds[threadIdx.x * V + i]);
-  }
-  __syncthreads();
-  #pragma unroll
-  for (int i = 0; i < V; ++i) lds[threadIdx.x * V + i] = a1[i];
-  __syncthreads();
-  for (int s = rpb / 2; s > 0; s >>= 1) {
-    if (row_in_block < s) {
+      lds[threadIdx.x * V + i] = pass ? a1[i] : a0[i];
+    __syncthreads();
+    for (int s = rpb / 2; s > 0; s >>= 1) {
+      if (row_in_block < s) {
+        #pragma unroll
+        for (int i = 0; i < V; ++i)
+          lds[threadIdx.x * V + i] +=
+              lds[(threadIdx.x + s * tpr) * V + i];
+      }
+      __syncthreads();
+    }
+    if (row_in_block == 0) {
       #pragma unroll
       for (int i = 0; i < V; ++i)
-        lds[threadIdx.x * V + i] += lds[(threadIdx.x + s * tpr) * V + i];
+        partial[pass * nbC + (long)blockIdx.x * C + c0 + i] =
+            lds[threadIdx.x * V + i];
     }
     __syncthreads();
-  }
-  if (row_in_block == 0) {
-    #pragma unroll
-    for (int i = 0; i < V; ++i)
-      atomicAdd(&dxhat_sum[c0 + i], lds[threadIdx.x * V + i]);
   }
 }
 
 // --------------------------------------------------------------------
-// backward dx: dx = gamma*invstd * (dy_eff - dsum/R - xhat*dxhat_sum/R)
-// optional dres = dy_eff (gradient of the fused residual input)
+// backward stage 2: fold partials -> dgamma/dbeta and the three
+// per-channel dx coefficients:
+//   dx = P*dy_eff + Q*x + S
+//   P = gamma*invstd
+//   Q = -P * invstd * (dxhat_sum/R)
+//   S = -P * (dsum/R) - Q * mean
+// --------------------------------------------------------------------
+__global__ void bn_bwd_fold_kernel(
+    const float* __restrict__ partial, int nblocks,
+    const float* __restrict__ gamma, const float* __restrict__ mean,
+    const float* __restrict__ invstd, float* __restrict__ dgamma,
+    float* __restrict__ dbeta, float* __restrict__ coefP,
+    float* __restrict__ coefQ, float* __restrict__ coefS, long R,
+    int C) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const long nbC = (long)nblocks * C;
+  float dsum = 0.f, dxhat = 0.f;
+  for (int b = 0; b < nblocks; ++b) {
+    dsum += partial[(long)b * C + c];
+    dxhat += partial[nbC + (long)b * C + c];
+  }
+  dbeta[c] = dsum;
+  dgamma[c] = dxhat;
+  const float invR = 1.f / (float)R;
+  const float P = gamma[c] * invstd[c];
+  const float Q = -P * invstd[c] * dxhat * invR;
+  coefP[c] = P;
+  coefQ[c] = Q;
+  coefS[c] = -P * dsum * invR - Q * mean[c];
+}
+
+// --------------------------------------------------------------------
+// backward dx elementwise: dx = P*dy_eff + Q*x + S [, dres = dy_eff]
 // --------------------------------------------------------------------
 template <typename T, bool RELU, bool RES>
 __global__ __launch_bounds__(kT) void bn_bwd_dx_kernel(
     const T* __restrict__ dy, const T* __restrict__ y,
-    const T* __restrict__ x, const float* __restrict__ mean,
-    const float* __restrict__ invstd, const float* __restrict__ gamma,
-    const float* __restrict__ dsum, const float* __restrict__ dxhat_sum,
-    T* __restrict__ dx, T* __restrict__ dres, long total, int C,
-    float invR) {
+    const T* __restrict__ x, const float* __restrict__ coefP,
+    const float* __restrict__ coefQ, const float* __restrict__ coefS,
+    T* __restrict__ dx, T* __restrict__ dres, long total, int C) {
   constexpr int V = BnVec<T>::V;
   const long stride = (long)gridDim.x * kT;
   const long nvec = total / V;
@@ -307,27 +339,26 @@ __global__ __launch_bounds__(kT) void bn_bwd_dx_kernel(
     load_vec<T, V>(xv, x + idx);
     T yv[V];
     if constexpr (RELU) load_vec<T, V>(yv, y + idx);
-    T dxv[V];
-    T drv[V];
+    float P[V], Q[V], S[V];
+    load_coef<V>(P, coefP + c0);
+    load_coef<V>(Q, coefQ + c0);
+    load_coef<V>(S, coefS + c0);
+    T dxv[V], drv[V];
     #pragma unroll
     for (int i = 0; i < V; ++i) {
-      const int c = c0 + i;
       float d = bn_tof<T>(dv[i]);
       if constexpr (RELU) d = bn_tof<T>(yv[i]) > 0.f ? d : 0.f;
       if constexpr (RES) drv[i] = bn_fromf<T>(d);
-      const float xh = (bn_tof<T>(xv[i]) - mean[c]) * invstd[c];
-      const float g = gamma[c] * invstd[c] *
-          (d - dsum[c] * invR - xh * dxhat_sum[c] * invR);
-      dxv[i] = bn_fromf<T>(g);
+      dxv[i] = bn_fromf<T>(P[i] * d + Q[i] * bn_tof<T>(xv[i]) + S[i]);
     }
     store_vec<T, V>(dx + idx, dxv);
     if constexpr (RES) store_vec<T, V>(dres + idx, drv);
   }
 }
 
-long bn_grid_rows(long R, int rpb) {
+int bn_grid_rows(long R, int rpb) {
   long blocks = (R + rpb - 1) / rpb;
-  return std::min<long>(std::max<long>(blocks, 1), 8192);
+  return (int)std::min<long>(std::max<long>(blocks, 1), kMaxStage1);
 }
 
 long bn_grid_elems(long nvec) {
@@ -347,19 +378,18 @@ void bn_fwd_impl(const torch::Tensor& x, torch::Tensor& y,
   auto scale = torch::empty({C}, opts);
   auto bias = torch::empty({C}, opts);
   if (training) {
-    auto sum = torch::zeros({C}, opts);
-    auto sumsq = torch::zeros({C}, opts);
     const int tpr = C / BnVec<T>::V;
     const int rpb = kT / tpr;
-    hipLaunchKernelGGL((bn_stats_kernel<T>),
-                       dim3(bn_grid_rows(R, rpb)), dim3(kT), 0, stream,
+    const int nb = bn_grid_rows(R, rpb);
+    auto partial = torch::empty({2, nb, C}, opts);
+    hipLaunchKernelGGL((bn_stats_kernel<T>), dim3(nb), dim3(kT), 0,
+                       stream,
                        reinterpret_cast<const T*>(x.data_ptr()),
-                       sum.data_ptr<float>(), sumsq.data_ptr<float>(), R,
-                       C);
+                       partial.data_ptr<float>(), R, C);
     CHECK_HIP_BN(hipGetLastError());
-    hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256),
-                       dim3(256), 0, stream, sum.data_ptr<float>(),
-                       sumsq.data_ptr<float>(), gamma.data_ptr<float>(),
+    hipLaunchKernelGGL(bn_stats_fold_kernel, dim3((C + 255) / 256),
+                       dim3(256), 0, stream, partial.data_ptr<float>(),
+                       nb, gamma.data_ptr<float>(),
                        beta.data_ptr<float>(),
                        running_mean.defined()
                            ? running_mean.data_ptr<float>() : nullptr,
@@ -408,37 +438,47 @@ void bn_bwd_impl(const torch::Tensor& dy, const torch::Tensor& y,
                  long R, int C, hipStream_t stream) {
   const int tpr = C / BnVec<T>::V;
   const int rpb = kT / tpr;
-  // dgamma/dbeta double as the reduce accumulators (zeroed by caller)
+  const int nb = bn_grid_rows(R, rpb);
+  auto fopts = gamma.options().dtype(at::kFloat);
+  auto partial = torch::empty({2, nb, C}, fopts);
   #define RED(RELU_)                                                     \
-    hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, RELU_>),                 \
-                       dim3(bn_grid_rows(R, rpb)), dim3(kT), 0, stream,  \
+    hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, RELU_>), dim3(nb),       \
+                       dim3(kT), 0, stream,                              \
                        reinterpret_cast<const T*>(dy.data_ptr()),        \
                        reinterpret_cast<const T*>(y.data_ptr()),         \
                        reinterpret_cast<const T*>(x.data_ptr()),         \
                        mean.data_ptr<float>(), invstd.data_ptr<float>(), \
-                       dbeta.data_ptr<float>(),                          \
-                       dgamma.data_ptr<float>(), R, C)
+                       partial.data_ptr<float>(), R, C)
   if (relu) RED(true); else RED(false);
   #undef RED
+  CHECK_HIP_BN(hipGetLastError());
+
+  auto coefP = torch::empty({C}, fopts);
+  auto coefQ = torch::empty({C}, fopts);
+  auto coefS = torch::empty({C}, fopts);
+  hipLaunchKernelGGL(bn_bwd_fold_kernel, dim3((C + 255) / 256),
+                     dim3(256), 0, stream, partial.data_ptr<float>(), nb,
+                     gamma.data_ptr<float>(), mean.data_ptr<float>(),
+                     invstd.data_ptr<float>(), dgamma.data_ptr<float>(),
+                     dbeta.data_ptr<float>(), coefP.data_ptr<float>(),
+                     coefQ.data_ptr<float>(), coefS.data_ptr<float>(), R,
+                     C);
   CHECK_HIP_BN(hipGetLastError());
 
   const long total = R * (long)C;
   const long grid = bn_grid_elems(total / BnVec<T>::V);
   T* dresp = dres.has_value()
       ? reinterpret_cast<T*>(dres->data_ptr()) : nullptr;
-  const float invR = 1.f / (float)R;
   #define DX(RELU_, RES_)                                                \
     hipLaunchKernelGGL((bn_bwd_dx_kernel<T, RELU_, RES_>), dim3(grid),   \
                        dim3(kT), 0, stream,                              \
                        reinterpret_cast<const T*>(dy.data_ptr()),        \
                        reinterpret_cast<const T*>(y.data_ptr()),         \
                        reinterpret_cast<const T*>(x.data_ptr()),         \
-                       mean.data_ptr<float>(), invstd.data_ptr<float>(), \
-                       gamma.data_ptr<float>(),                          \
-                       dbeta.data_ptr<float>(),                          \
-                       dgamma.data_ptr<float>(), \
+                       coefP.data_ptr<float>(), coefQ.data_ptr<float>(), \
+                       coefS.data_ptr<float>(),                          \
                        reinterpret_cast<T*>(dx.data_ptr()), dresp,       \
-                       total, C, invR)
+                       total, C)
   if (relu && dresp) DX(true, true);
   else if (relu) DX(true, false);
   else if (dresp) DX(false, true);
@@ -498,8 +538,8 @@ std::vector<torch::Tensor> fused_bn_bwd(
   auto stream = at::hip::getCurrentHIPStream().stream();
   auto dx = torch::empty_like(x);
   auto fopts = gamma.options().dtype(at::kFloat);
-  auto dgamma = torch::zeros({C}, fopts);
-  auto dbeta = torch::zeros({C}, fopts);
+  auto dgamma = torch::empty({C}, fopts);
+  auto dbeta = torch::empty({C}, fopts);
   c10::optional<torch::Tensor> dres;
   if (has_res) dres = torch::empty_like(x);
   if (x.scalar_type() == at::kBFloat16)
