@@ -100,6 +100,7 @@ class NativeDDP(nn.Module):
         self._hooks = []
         self._buckets: List[_Bucket] = []
         self._param_to_bucket: Dict[nn.Parameter, _Bucket] = {}
+        self._next_bucket = 0
 
         self._sync_initial_state()
         self._build_buckets()
@@ -155,7 +156,24 @@ class NativeDDP(nn.Module):
         bucket = self._param_to_bucket[param]
         bucket.ready.add(param)
         if len(bucket.ready) == len(bucket.params):
-            self._launch_bucket(bucket)
+            # RCCL requires every rank to issue collectives in the SAME
+            # order; readiness order can diverge across ranks (dynamic
+            # control flow), so buckets launch strictly in index order —
+            # a ready-but-early bucket waits for its predecessors
+            # (torch DDP enforces the same invariant).
+            self._maybe_launch_in_order()
+
+    def _maybe_launch_in_order(self) -> None:
+        while self._next_bucket < len(self._buckets):
+            b = self._buckets[self._next_bucket]
+            if b.reduced:
+                self._next_bucket += 1
+                continue
+            if len(b.ready) == len(b.params):
+                self._launch_bucket(b)
+                self._next_bucket += 1
+            else:
+                break
 
     def _launch_bucket(self, bucket: _Bucket) -> None:
         from .. import ops
@@ -194,6 +212,7 @@ class NativeDDP(nn.Module):
                 p.grad = view
             bucket.ready.clear()
             bucket.reduced = False
+        self._next_bucket = 0
 
     @contextlib.contextmanager
     def no_sync(self):

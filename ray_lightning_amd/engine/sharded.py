@@ -208,6 +208,7 @@ class ShardedDDP(nn.Module):
         self.average = average
         self._sync_enabled = True
         self._hooks = []
+        self._next_bucket = 0
 
         cap = int(bucket_cap_mb * 1024 * 1024)
         self._buckets = []  # (owner, params, flat, offsets, ready)
@@ -262,7 +263,22 @@ class ShardedDDP(nn.Module):
         b = self._param_to_bucket[param]
         b["ready"].add(param)
         if len(b["ready"]) == len(b["params"]):
-            self._launch(b)
+            # launch strictly in bucket-index order: RCCL requires a
+            # uniform collective issue order across ranks (see
+            # NativeDDP._maybe_launch_in_order).
+            self._maybe_launch_in_order()
+
+    def _maybe_launch_in_order(self) -> None:
+        while self._next_bucket < len(self._buckets):
+            b = self._buckets[self._next_bucket]
+            if b["reduced"]:
+                self._next_bucket += 1
+                continue
+            if len(b["ready"]) == len(b["params"]):
+                self._launch(b)
+                self._next_bucket += 1
+            else:
+                break
 
     def _launch(self, b) -> None:
         from .. import ops
@@ -305,6 +321,7 @@ class ShardedDDP(nn.Module):
                     p.grad = None
             b["ready"].clear()
             b["reduced"] = False
+        self._next_bucket = 0
 
     @contextlib.contextmanager
     def no_sync(self):
