@@ -173,8 +173,12 @@ class RayStrategy(Strategy):
     # ------------------------------------------------------------------ #
     def _worker_setup(self, process_idx: int) -> None:
         from ..trainer.trainer import _reset_seed
+        from ..util import rank_zero_only
         _reset_seed()
         self.set_world_ranks(process_idx)
+        # "rank-0-only" helpers must act on the ACTOR's rank
+        # (reference ray_ddp.py:169)
+        rank_zero_only.rank = self.global_rank
         if self.world_size > 1:
             init_control_plane(self.global_rank, self.world_size)
             self._comm = TorchDistCommunicator()

@@ -118,3 +118,35 @@ def find_free_port() -> int:
         s.bind(("", 0))
         s.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
         return s.getsockname()[1]
+
+
+# --------------------------------------------------------------------- #
+# rank-zero utilities (reference re-points rank_zero_only.rank inside
+# every worker, ray_ddp.py:169, so "rank-0-only" user logging works in
+# actors)
+# --------------------------------------------------------------------- #
+def rank_zero_only(fn: Callable) -> Callable:
+    """Decorator: run ``fn`` only on the global rank-0 worker."""
+    import functools
+
+    @functools.wraps(fn)
+    def wrapped(*args, **kwargs):
+        if getattr(rank_zero_only, "rank", 0) == 0:
+            return fn(*args, **kwargs)
+        return None
+
+    return wrapped
+
+
+rank_zero_only.rank = 0  # re-pointed per worker by _worker_setup
+
+
+@rank_zero_only
+def rank_zero_info(*args, **kwargs) -> None:
+    print(*args, **kwargs)
+
+
+@rank_zero_only
+def rank_zero_warn(*args, **kwargs) -> None:
+    import warnings
+    warnings.warn(" ".join(str(a) for a in args))
