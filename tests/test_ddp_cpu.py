@@ -206,3 +206,23 @@ def test_driver_model_gets_trained_weights(tmp_path):
     after = torch.cat([p.flatten() for p in model.parameters()])
     assert not torch.equal(before, after)
     assert trainer.model is model
+
+
+class _FailingModel(BoringModel):
+    def training_step(self, batch, batch_idx):
+        if batch_idx == 2:
+            raise RuntimeError("intentional-worker-failure")
+        return super().training_step(batch, batch_idx)
+
+
+def test_worker_failure_fate_sharing(tmp_path):
+    """A worker exception aborts the driver fit with the original error
+    and tears the workers down (reference util.py:63-65 +
+    ray_launcher.py:116-128)."""
+    from ray_lightning_amd.runtime.actor import RemoteError
+    model = _FailingModel()
+    strategy = RayStrategy(num_workers=2)
+    trainer = get_trainer(str(tmp_path), strategy=strategy)
+    with pytest.raises(RemoteError, match="intentional-worker-failure"):
+        trainer.fit(model)
+    assert strategy.launcher._workers == []  # torn down, no leaks
