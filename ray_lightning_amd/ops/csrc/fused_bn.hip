@@ -132,30 +132,31 @@ __global__ __launch_bounds__(kT) void bn_stats_kernel(
 }
 
 // --------------------------------------------------------------------
-// stage 2: fold partials -> mean/invstd + scale/bias (+ running)
-// Parallel fold: each block owns kFoldC channels; the 256 threads are
-// arranged (kFoldC channels x 256/kFoldC partial-lanes) so reads of
-// partial[b*C + c] are coalesced in c and the nblocks loop is split
-// across lanes (a v2 used one thread per channel: <=2048 threads on
-// the whole chip, 270us latency-bound vs ~20us for stage 1).
+// stage 2a: fold the [nblocks, C] partials down to [kB2, C] with a 2-D
+// grid (channel-groups x b-slices) so even C=64 layers get enough
+// workgroups to hide memory latency (a v3 used one block per 64
+// channels: 1 block on the whole chip for layer1, 74us latency-bound).
+// Deterministic: fixed slice boundaries, no atomics.
+// stage 2b: tiny per-channel kernel folds the kB2 rows and computes
+// the coefficients.
 // --------------------------------------------------------------------
-constexpr int kFoldC = 64;
+constexpr int kFoldC = 64;   // channels per fold block
+constexpr int kB2 = 16;      // second-level partial rows
 
-__global__ __launch_bounds__(kT) void bn_stats_fold_kernel(
+__global__ __launch_bounds__(kT) void bn_fold2_kernel(
     const float* __restrict__ partial, int nblocks,
-    const float* __restrict__ gamma, const float* __restrict__ beta,
-    float* __restrict__ running_mean, float* __restrict__ running_var,
-    float* __restrict__ mean_out, float* __restrict__ invstd_out,
-    float* __restrict__ scale_out, float* __restrict__ bias_out,
-    long R, int C, float momentum, float eps) {
+    float* __restrict__ partial2, int C) {
+  // grid: (ceil(C/kFoldC), kB2); threads (kFoldC x lanes)
   const int lanes = kT / kFoldC;
   const int c = blockIdx.x * kFoldC + threadIdx.x % kFoldC;
   const int lane = threadIdx.x / kFoldC;
-  const bool active = c < C;  // keep barriers uniform on the C tail
+  const bool active = c < C;
+  const int b_begin = (int)(((long)blockIdx.y * nblocks) / kB2);
+  const int b_end = (int)(((long)(blockIdx.y + 1) * nblocks) / kB2);
   const long nbC = (long)nblocks * C;
   float s = 0.f, s2 = 0.f;
   if (active)
-    for (int b = lane; b < nblocks; b += lanes) {
+    for (int b = b_begin + lane; b < b_end; b += lanes) {
       s += partial[(long)b * C + c];
       s2 += partial[nbC + (long)b * C + c];
     }
@@ -171,8 +172,27 @@ __global__ __launch_bounds__(kT) void bn_stats_fold_kernel(
     __syncthreads();
   }
   if (lane != 0 || !active) return;
-  s = lsum[threadIdx.x];
-  s2 = lsq[threadIdx.x];
+  const long b2C = (long)kB2 * C;
+  partial2[(long)blockIdx.y * C + c] = lsum[threadIdx.x];
+  partial2[b2C + (long)blockIdx.y * C + c] = lsq[threadIdx.x];
+}
+
+__global__ void bn_stats_coef_kernel(
+    const float* __restrict__ partial2,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    float* __restrict__ running_mean, float* __restrict__ running_var,
+    float* __restrict__ mean_out, float* __restrict__ invstd_out,
+    float* __restrict__ scale_out, float* __restrict__ bias_out,
+    long R, int C, float momentum, float eps) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const long b2C = (long)kB2 * C;
+  float s = 0.f, s2 = 0.f;
+  #pragma unroll
+  for (int b = 0; b < kB2; ++b) {
+    s += partial2[(long)b * C + c];
+    s2 += partial2[b2C + (long)b * C + c];
+  }
   const float m = s / (float)R;
   const float var = fmaxf(s2 / (float)R - m * m, 0.f);
   const float inv = rsqrtf(var + eps);
@@ -317,38 +337,22 @@ __global__ __launch_bounds__(kT) void bn_bwd_reduce_kernel(
 //   Q = -P * invstd * (dxhat_sum/R)
 //   S = -P * (dsum/R) - Q * mean
 // --------------------------------------------------------------------
-__global__ __launch_bounds__(kT) void bn_bwd_fold_kernel(
-    const float* __restrict__ partial, int nblocks,
+__global__ void bn_bwd_coef_kernel(
+    const float* __restrict__ partial2,
     const float* __restrict__ gamma, const float* __restrict__ mean,
     const float* __restrict__ invstd, float* __restrict__ dgamma,
     float* __restrict__ dbeta, float* __restrict__ coefP,
     float* __restrict__ coefQ, float* __restrict__ coefS, long R,
     int C) {
-  const int lanes = kT / kFoldC;
-  const int c = blockIdx.x * kFoldC + threadIdx.x % kFoldC;
-  const int lane = threadIdx.x / kFoldC;
-  const bool active = c < C;  // keep barriers uniform on the C tail
-  const long nbC = (long)nblocks * C;
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const long b2C = (long)kB2 * C;
   float dsum = 0.f, dxhat = 0.f;
-  if (active)
-    for (int b = lane; b < nblocks; b += lanes) {
-      dsum += partial[(long)b * C + c];
-      dxhat += partial[nbC + (long)b * C + c];
-    }
-  __shared__ float lsum[kT], lsq[kT];
-  lsum[threadIdx.x] = dsum;
-  lsq[threadIdx.x] = dxhat;
-  __syncthreads();
-  for (int st = lanes / 2; st > 0; st >>= 1) {
-    if (lane < st) {
-      lsum[threadIdx.x] += lsum[threadIdx.x + st * kFoldC];
-      lsq[threadIdx.x] += lsq[threadIdx.x + st * kFoldC];
-    }
-    __syncthreads();
+  #pragma unroll
+  for (int b = 0; b < kB2; ++b) {
+    dsum += partial2[(long)b * C + c];
+    dxhat += partial2[b2C + (long)b * C + c];
   }
-  if (lane != 0 || !active) return;
-  dsum = lsum[threadIdx.x];
-  dxhat = lsq[threadIdx.x];
   dbeta[c] = dsum;
   dgamma[c] = dxhat;
   const float invR = 1.f / (float)R;
@@ -423,14 +427,20 @@ void bn_fwd_impl(const torch::Tensor& x, torch::Tensor& y,
     const int rpb = kT / tpr;
     const int nb = bn_grid_rows(R, rpb);
     auto partial = torch::empty({2, nb, C}, opts);
+    auto partial2 = torch::empty({2, kB2, C}, opts);
     hipLaunchKernelGGL((bn_stats_kernel<T>), dim3(nb), dim3(kT), 0,
                        stream,
                        reinterpret_cast<const T*>(x.data_ptr()),
                        partial.data_ptr<float>(), R, C);
     CHECK_HIP_BN(hipGetLastError());
-    hipLaunchKernelGGL(bn_stats_fold_kernel, dim3((C + 63) / 64),
-                       dim3(kT), 0, stream, partial.data_ptr<float>(),
-                       nb, gamma.data_ptr<float>(),
+    hipLaunchKernelGGL(bn_fold2_kernel,
+                       dim3((C + kFoldC - 1) / kFoldC, kB2), dim3(kT), 0,
+                       stream, partial.data_ptr<float>(), nb,
+                       partial2.data_ptr<float>(), C);
+    CHECK_HIP_BN(hipGetLastError());
+    hipLaunchKernelGGL(bn_stats_coef_kernel, dim3((C + 255) / 256),
+                       dim3(256), 0, stream, partial2.data_ptr<float>(),
+                       gamma.data_ptr<float>(),
                        beta.data_ptr<float>(),
                        running_mean.defined()
                            ? running_mean.data_ptr<float>() : nullptr,
@@ -482,6 +492,7 @@ void bn_bwd_impl(const torch::Tensor& dy, const torch::Tensor& y,
   const int nb = bn_grid_rows(R, rpb);
   auto fopts = gamma.options().dtype(at::kFloat);
   auto partial = torch::empty({2, nb, C}, fopts);
+  auto partial2 = torch::empty({2, kB2, C}, fopts);
   #define RED(RELU_)                                                     \
     hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, RELU_>), dim3(nb),       \
                        dim3(kT), 0, stream,                              \
@@ -497,8 +508,13 @@ void bn_bwd_impl(const torch::Tensor& dy, const torch::Tensor& y,
   auto coefP = torch::empty({C}, fopts);
   auto coefQ = torch::empty({C}, fopts);
   auto coefS = torch::empty({C}, fopts);
-  hipLaunchKernelGGL(bn_bwd_fold_kernel, dim3((C + 63) / 64),
-                     dim3(kT), 0, stream, partial.data_ptr<float>(), nb,
+  hipLaunchKernelGGL(bn_fold2_kernel,
+                     dim3((C + kFoldC - 1) / kFoldC, kB2), dim3(kT), 0,
+                     stream, partial.data_ptr<float>(), nb,
+                     partial2.data_ptr<float>(), C);
+  CHECK_HIP_BN(hipGetLastError());
+  hipLaunchKernelGGL(bn_bwd_coef_kernel, dim3((C + 255) / 256),
+                     dim3(256), 0, stream, partial2.data_ptr<float>(),
                      gamma.data_ptr<float>(), mean.data_ptr<float>(),
                      invstd.data_ptr<float>(), dgamma.data_ptr<float>(),
                      dbeta.data_ptr<float>(), coefP.data_ptr<float>(),
