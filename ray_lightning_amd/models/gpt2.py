@@ -120,13 +120,16 @@ class GPT2(nn.Module):
         for block in self.h:
             x = block(x)
         x = self.ln_f(x)
-        logits = self.lm_head(x)
-        loss = None
         if targets is not None:
-            loss = F.cross_entropy(
-                logits.view(-1, logits.size(-1)).float(),
+            # chunked LM-head + CE: never materializes the [B*T, V]
+            # logits (ops/chunked_ce.py); identical numerics to
+            # F.cross_entropy(logits.float(), t)
+            from ..ops.chunked_ce import chunked_cross_entropy
+            loss = chunked_cross_entropy(
+                x.reshape(-1, x.size(-1)), self.lm_head.weight,
                 targets.reshape(-1))
-        return logits, loss
+            return None, loss
+        return self.lm_head(x), None
 
 
 def gpt2_xl() -> GPT2:
