@@ -525,16 +525,55 @@ __global__ __launch_bounds__(kThreads) void sharded_adam_kernel(
   #pragma unroll
   for (int u = 0; u < UNROLL; ++u) {
     const long idx = base + ((long)u * kThreads + threadIdx.x) * 4;
-    for (long i = idx; i < numel && i < idx + 4; ++i) {
-      float gi = ToFloat<GradT>::conv(g[i]);
-      float pi = master[i];
-      if (adamw) pi *= (1.f - lr * weight_decay);
-      else gi += weight_decay * pi;
-      m[i] = beta1 * m[i] + (1.f - beta1) * gi;
-      v[i] = beta2 * v[i] + (1.f - beta2) * gi * gi;
-      pi -= lr * (m[i] * inv_bc1) / (sqrtf(v[i]) * inv_sqrt_bc2 + eps);
-      master[i] = pi;
-      p16[i] = __float2bfloat16(pi);
+    if (idx + 4 <= numel) {
+      // vectorized: float4 state, 4-lane grads/params (a scalar v1 of
+      // this loop was 3x off the bandwidth bound on GPT-2-XL)
+      float4 mstv = *reinterpret_cast<float4*>(master + idx);
+      float4 mv = *reinterpret_cast<float4*>(m + idx);
+      float4 vv = *reinterpret_cast<float4*>(v + idx);
+      GradT gv[4];
+      if constexpr (sizeof(GradT) == 2)
+        *reinterpret_cast<int2*>(gv) =
+            *reinterpret_cast<const int2*>(g + idx);
+      else
+        *reinterpret_cast<int4*>(gv) =
+            *reinterpret_cast<const int4*>(g + idx);
+      float pe[4] = {mstv.x, mstv.y, mstv.z, mstv.w};
+      float me[4] = {mv.x, mv.y, mv.z, mv.w};
+      float ve[4] = {vv.x, vv.y, vv.z, vv.w};
+      __hip_bfloat16 out16[4];
+      #pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        float gi = ToFloat<GradT>::conv(gv[i]);
+        if (adamw) pe[i] *= (1.f - lr * weight_decay);
+        else gi += weight_decay * pe[i];
+        me[i] = beta1 * me[i] + (1.f - beta1) * gi;
+        ve[i] = beta2 * ve[i] + (1.f - beta2) * gi * gi;
+        pe[i] -= lr * (me[i] * inv_bc1) /
+                 (sqrtf(ve[i]) * inv_sqrt_bc2 + eps);
+        out16[i] = __float2bfloat16(pe[i]);
+      }
+      *reinterpret_cast<float4*>(master + idx) =
+          float4{pe[0], pe[1], pe[2], pe[3]};
+      *reinterpret_cast<float4*>(m + idx) =
+          float4{me[0], me[1], me[2], me[3]};
+      *reinterpret_cast<float4*>(v + idx) =
+          float4{ve[0], ve[1], ve[2], ve[3]};
+      *reinterpret_cast<int2*>(p16 + idx) =
+          *reinterpret_cast<const int2*>(out16);
+    } else {
+      for (long i = idx; i < numel && i < idx + 4; ++i) {
+        float gi = ToFloat<GradT>::conv(g[i]);
+        float pi = master[i];
+        if (adamw) pi *= (1.f - lr * weight_decay);
+        else gi += weight_decay * pi;
+        m[i] = beta1 * m[i] + (1.f - beta1) * gi;
+        v[i] = beta2 * v[i] + (1.f - beta2) * gi * gi;
+        pi -= lr * (m[i] * inv_bc1) /
+              (sqrtf(v[i]) * inv_sqrt_bc2 + eps);
+        master[i] = pi;
+        p16[i] = __float2bfloat16(pi);
+      }
     }
   }
 }
