@@ -120,8 +120,8 @@ def main() -> None:
         cfg = (GPT2Config.gpt2_xl() if args.model == "gpt2-xl"
                else GPT2Config.gpt2())
         cfg.n_positions = max(args.seq_len, 1024)
-        model = GPT2(cfg).to(device)
-        model = model.to(torch.bfloat16)
+        from ray_lightning_amd.models.gpt2 import to_bf16_training
+        model = to_bf16_training(GPT2(cfg).to(device))
         model.train()
         decay = [pm for pm in model.parameters() if pm.dim() >= 2]
         nodecay = [pm for pm in model.parameters() if pm.dim() < 2]

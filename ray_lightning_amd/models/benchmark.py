@@ -123,7 +123,8 @@ class GPT2LM(LightningModule):
         cfg.n_positions = max(seq_len, 1024)
         self.model = GPT2(cfg)
         if bf16_weights:
-            self.model = self.model.to(torch.bfloat16)
+            from .gpt2 import to_bf16_training
+            self.model = to_bf16_training(self.model)
         self.lr = lr
         self.weight_decay = weight_decay
         self.batch_size = batch_size

@@ -16,6 +16,8 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from ..ops.fused_ln import FusedLayerNorm
+
 
 @dataclass
 class GPT2Config:
@@ -76,12 +78,14 @@ class MLP(nn.Module):
 class Block(nn.Module):
     def __init__(self, cfg: GPT2Config):
         super().__init__()
-        self.ln_1 = nn.LayerNorm(cfg.n_embd)
+        self.ln_1 = FusedLayerNorm(cfg.n_embd)
         self.attn = CausalSelfAttention(cfg)
-        self.ln_2 = nn.LayerNorm(cfg.n_embd)
+        self.ln_2 = FusedLayerNorm(cfg.n_embd)
         self.mlp = MLP(cfg)
 
     def forward(self, x):
+        # standalone (unfused) path; GPT2.forward drives the fused
+        # residual+LN chain across block boundaries instead
         x = x + self.attn(self.ln_1(x))
         x = x + self.mlp(self.ln_2(x))
         return x
@@ -94,7 +98,7 @@ class GPT2(nn.Module):
         self.wte = nn.Embedding(cfg.vocab_size, cfg.n_embd)
         self.wpe = nn.Embedding(cfg.n_positions, cfg.n_embd)
         self.h = nn.ModuleList(Block(cfg) for _ in range(cfg.n_layer))
-        self.ln_f = nn.LayerNorm(cfg.n_embd)
+        self.ln_f = FusedLayerNorm(cfg.n_embd)
         self.lm_head = nn.Linear(cfg.n_embd, cfg.vocab_size, bias=False)
         self.lm_head.weight = self.wte.weight  # weight tying
 
@@ -117,9 +121,18 @@ class GPT2(nn.Module):
         B, T = idx.shape
         pos = torch.arange(T, device=idx.device)
         x = self.wte(idx) + self.wpe(pos)
-        for block in self.h:
-            x = block(x)
-        x = self.ln_f(x)
+        # fused pre-norm chain: each residual add is folded into the
+        # NEXT LayerNorm's forward (ops/fused_ln.py), crossing block
+        # boundaries; h is always ln(x) of the current position.
+        h = self.h[0].ln_1(x)
+        for i, block in enumerate(self.h):
+            attn_out = block.attn(h)
+            x, h = block.ln_2(attn_out, residual=x)
+            mlp_out = block.mlp(h)
+            next_ln = (self.h[i + 1].ln_1 if i + 1 < len(self.h)
+                       else self.ln_f)
+            x, h = next_ln(mlp_out, residual=x)
+        x = h  # = ln_f(x)
         if targets is not None:
             # chunked LM-head + CE: never materializes the [B*T, V]
             # logits (ops/chunked_ce.py); identical numerics to
@@ -130,6 +143,17 @@ class GPT2(nn.Module):
                 targets.reshape(-1))
             return None, loss
         return self.lm_head(x), None
+
+
+def to_bf16_training(model: nn.Module) -> nn.Module:
+    """bf16 weights with fp32 normalization params: the fused LN/BN
+    kernels keep their statistics math and affine params in fp32 (the
+    numerically sane split, and what engages the fused HIP path)."""
+    model = model.to(torch.bfloat16)
+    for m in model.modules():
+        if isinstance(m, (nn.LayerNorm,)):
+            m.float()
+    return model
 
 
 def gpt2_xl() -> GPT2:

@@ -638,6 +638,15 @@ std::vector<torch::Tensor> fused_bn_bwd(
     torch::Tensor mean, torch::Tensor invstd, torch::Tensor gamma,
     bool relu, bool has_res);
 
+// fused residual-add + LayerNorm — defined in fused_ln.hip
+std::vector<torch::Tensor> fused_ln_fwd(
+    torch::Tensor a, c10::optional<torch::Tensor> b,
+    torch::Tensor gamma, torch::Tensor beta, double eps);
+std::vector<torch::Tensor> fused_ln_bwd(
+    torch::Tensor dy, torch::Tensor x,
+    c10::optional<torch::Tensor> dext, torch::Tensor gamma,
+    torch::Tensor mean, torch::Tensor invstd);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("multi_tensor_pack", &multi_tensor_pack,
         "multi-tensor flatten/cast into bucket slots");
@@ -645,6 +654,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused NHWC BN(+ReLU)(+residual) forward");
   m.def("fused_bn_bwd", &fused_bn_bwd,
         "fused NHWC BN(+ReLU)(+residual) backward");
+  m.def("fused_ln_fwd", &fused_ln_fwd,
+        "fused residual-add + LayerNorm forward");
+  m.def("fused_ln_bwd", &fused_ln_bwd,
+        "fused residual-add + LayerNorm backward");
   m.def("scale_inplace", &scale_inplace, "flat *= s");
   m.def("scale_cast", &scale_cast, "dst_f32 = src_bf16 * s");
   m.def("fused_sgd", &fused_sgd, "fused multi-tensor SGD(momentum)");
