@@ -59,8 +59,11 @@ class _ChunkedLinearCE(torch.autograd.Function):
 
 def chunked_cross_entropy(x: torch.Tensor, weight: torch.Tensor,
                           targets: torch.Tensor,
-                          chunk_rows: int = 1024) -> torch.Tensor:
+                          chunk_rows: int = None) -> torch.Tensor:
     """Mean cross-entropy of ``x @ weight.T`` against ``targets``
     without materializing the full logits. ``x``: [N, C] (bf16/f32),
     ``weight``: [V, C], ``targets``: [N] int64."""
+    if chunk_rows is None:
+        import os
+        chunk_rows = int(os.environ.get("RLA_CE_CHUNK_ROWS", "2048"))
     return _ChunkedLinearCE.apply(x, weight, targets, chunk_rows)
