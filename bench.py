@@ -51,7 +51,10 @@ def build_comm(rank: int, world: int, device: torch.device):
     torch-dist RCCL)."""
     init_control_plane(rank, world)
     control = TorchDistCommunicator()
-    if device.type == "cuda":
+    # PL_TORCH_DISTRIBUTED_BACKEND=gloo forces the CPU data plane (the
+    # fractional-GPU / ranks-sharing-one-device case, like the strategy)
+    if device.type == "cuda" and \
+            os.environ.get("PL_TORCH_DISTRIBUTED_BACKEND") != "gloo":
         from ray_lightning_amd.engine.rccl import (NativeRcclCommunicator,
                                                    rccl_available)
         if rccl_available():
