@@ -65,5 +65,5 @@ def chunked_cross_entropy(x: torch.Tensor, weight: torch.Tensor,
     ``weight``: [V, C], ``targets``: [N] int64."""
     if chunk_rows is None:
         import os
-        chunk_rows = int(os.environ.get("RLA_CE_CHUNK_ROWS", "2048"))
+        chunk_rows = int(os.environ.get("RLA_CE_CHUNK_ROWS", "8192"))
     return _ChunkedLinearCE.apply(x, weight, targets, chunk_rows)
