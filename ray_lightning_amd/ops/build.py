@@ -31,7 +31,9 @@ def build(verbose: bool = False) -> None:
     for name, sources, ldflags in (
         ("_hip_ops", [os.path.join(_CSRC, "hip_ops.hip"),
                       os.path.join(_CSRC, "fused_bn.hip"),
-                      os.path.join(_CSRC, "fused_ln.hip")], []),
+                      os.path.join(_CSRC, "fused_ln.hip"),
+                      os.path.join(_CSRC, "flash_attn.hip"),
+                      os.path.join(_CSRC, "mfma_probe.hip")], []),
         ("_rccl_comm", [os.path.join(_CSRC, "rccl_comm.hip")],
          ["-L/opt/rocm/lib", "-lrccl"]),
     ):

@@ -638,6 +638,15 @@ std::vector<torch::Tensor> fused_bn_bwd(
     torch::Tensor mean, torch::Tensor invstd, torch::Tensor gamma,
     bool relu, bool has_res);
 
+// flash attention (causal, hs=64) — defined in flash_attn.hip
+std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q,
+                                          torch::Tensor k,
+                                          torch::Tensor v, double scale);
+std::vector<torch::Tensor> flash_attn_bwd(
+    torch::Tensor dout, torch::Tensor q, torch::Tensor k,
+    torch::Tensor v, torch::Tensor o, torch::Tensor lse, double scale);
+torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
+
 // fused residual-add + LayerNorm — defined in fused_ln.hip
 std::vector<torch::Tensor> fused_ln_fwd(
     torch::Tensor a, c10::optional<torch::Tensor> b,
@@ -658,6 +667,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused residual-add + LayerNorm forward");
   m.def("fused_ln_bwd", &fused_ln_bwd,
         "fused residual-add + LayerNorm backward");
+  m.def("flash_attn_fwd", &flash_attn_fwd,
+        "causal flash attention forward (hs=64, bf16, MFMA)");
+  m.def("flash_attn_bwd", &flash_attn_bwd,
+        "causal flash attention backward");
+  m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
   m.def("scale_inplace", &scale_inplace, "flat *= s");
   m.def("scale_cast", &scale_cast, "dst_f32 = src_bf16 * s");
   m.def("fused_sgd", &fused_sgd, "fused multi-tensor SGD(momentum)");
