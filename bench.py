@@ -100,7 +100,10 @@ def main() -> None:
     on_gpu = torch.cuda.is_available()
 
     if on_gpu:
-        device = torch.device("cuda", local_rank)
+        # modulo: lets N ranks share fewer devices (fractional-GPU /
+        # gloo validation); the 8-GPU deployment maps 1:1
+        device = torch.device("cuda",
+                              local_rank % torch.cuda.device_count())
         torch.cuda.set_device(device)
         batch = args.batch_size
     else:
