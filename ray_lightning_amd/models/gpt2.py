@@ -60,7 +60,10 @@ class CausalSelfAttention(nn.Module):
         q = q.view(B, T, self.n_head, hs).transpose(1, 2)
         k = k.view(B, T, self.n_head, hs).transpose(1, 2)
         v = v.view(B, T, self.n_head, hs).transpose(1, 2)
-        y = F.scaled_dot_product_attention(q, k, v, is_causal=True)
+        # hand-written MFMA flash kernels when the shape qualifies
+        # (causal bf16 hs=64); SDPA (AOTriton) otherwise
+        from ..ops.flash_attn import flash_attention
+        y = flash_attention(q, k, v)
         y = y.transpose(1, 2).contiguous().view(B, T, C)
         return self.c_proj(y)
 
