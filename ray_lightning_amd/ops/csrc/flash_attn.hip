@@ -480,7 +480,6 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_kernel(
   const int wrow0 = qm0 + wave * 16;
   const int col0 = lane & 15;
 
-  __shared__ __bf16 lds_vt[HS * BN];    // V^T [hs][kv]
   __shared__ __bf16 lds_kt[HS * BN];    // K^T [hs][kv]
   __shared__ __bf16 lds_ds[4][16 * BN]; // per-wave dS
 
@@ -505,23 +504,19 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_kernel(
   const float l2e = 1.4426950408889634f;
   const int kv_end = qm0 + BM;
   for (int kn0 = 0; kn0 < kv_end && kn0 < T; kn0 += BN) {
-    // stage V^T / K^T images
+    // stage the K^T image (dQ's B operand contracts over kv)
     {
       const int r = threadIdx.x >> 3;
       const int c0 = (threadIdx.x & 7) * 8;
       #pragma unroll
       for (int rep = 0; rep < 2; ++rep) {
         const int row = r + rep * 32;
-        bf16x8 t1, t2;
-        *reinterpret_cast<int4*>(&t1) = *reinterpret_cast<const int4*>(
-            v + (long)(kn0 + row) * HS + c0);
+        bf16x8 t2;
         *reinterpret_cast<int4*>(&t2) = *reinterpret_cast<const int4*>(
             k + (long)(kn0 + row) * HS + c0);
         #pragma unroll
-        for (int i = 0; i < 8; ++i) {
-          lds_vt[(c0 + i) * BN + row] = t1[i];
+        for (int i = 0; i < 8; ++i)
           lds_kt[(c0 + i) * BN + row] = t2[i];
-        }
       }
     }
     __syncthreads();
@@ -537,9 +532,9 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_kernel(
         bf16x8 bk = load_frag_rowmajor(k, kn0 + 16 * n, kk, lane, HS);
         s[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             qf[kk], bk, s[n], 0, 0, 0);
-        bf16x8 bv;
-        *reinterpret_cast<int4*>(&bv) = *reinterpret_cast<const int4*>(
-            lds_vt + (16 * n + col0) * BN + kk * 32 + (lane >> 4) * 8);
+        // dP = dO · V^T contracts over hs -> row-major V fragments
+        // (col = kv row, k = hs), like the S kernel's K fragments
+        bf16x8 bv = load_frag_rowmajor(v, kn0 + 16 * n, kk, lane, HS);
         dp[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             dof[kk], bv, dp[n], 0, 0, 0);
       }
