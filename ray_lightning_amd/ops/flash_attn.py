@@ -17,7 +17,7 @@ from . import _load_ext
 
 
 def _usable(q: torch.Tensor) -> bool:
-    if os.environ.get("RLA_FLASH", "1") == "0":
+    if os.environ.get("RLA_FLASH", "0") != "1":  # opt-in until it beats SDPA
         return False
     return (q.is_cuda and q.dtype == torch.bfloat16 and q.dim() == 4
             and q.shape[-1] == 64 and q.shape[2] % 64 == 0
