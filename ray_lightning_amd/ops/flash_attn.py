@@ -20,7 +20,7 @@ def _usable(q: torch.Tensor) -> bool:
     if os.environ.get("RLA_FLASH", "0") != "1":  # opt-in until it beats SDPA
         return False
     return (q.is_cuda and q.dtype == torch.bfloat16 and q.dim() == 4
-            and q.shape[-1] == 64 and q.shape[2] % 64 == 0
+            and q.shape[-1] == 64 and q.shape[2] % 128 == 0
             and _load_ext() is not None
             and hasattr(_load_ext(), "flash_attn_fwd"))
 

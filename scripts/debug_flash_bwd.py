@@ -55,6 +55,6 @@ def check(B, H, T):
 
 
 if __name__ == "__main__":
-    check(1, 1, 64)
     check(1, 1, 128)
     check(2, 3, 256)
+    check(1, 2, 512)

@@ -1,7 +1,10 @@
 """MFMA layout probe + flash attention numerics vs fp32 reference,
 plus a timing comparison against torch SDPA (AOTriton)."""
 import math
+import os
 import time
+
+os.environ["RLA_FLASH"] = "1"  # force the custom path under test
 
 import pytest
 import torch
@@ -35,7 +38,7 @@ def _ref_attention(q, k, v, scale):
     return p @ v.float()
 
 
-@pytest.mark.parametrize("B,H,T", [(2, 3, 256), (1, 2, 1024)])
+@pytest.mark.parametrize("B,H,T", [(2, 3, 256), (1, 2, 1024), (1, 1, 128)])
 def test_flash_fwd_numerics(B, H, T):
     from ray_lightning_amd import ops
     ext = ops._load_ext()
