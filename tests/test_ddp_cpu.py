@@ -226,3 +226,31 @@ def test_worker_failure_fate_sharing(tmp_path):
     with pytest.raises(RemoteError, match="intentional-worker-failure"):
         trainer.fit(model)
     assert strategy.launcher._workers == []  # torn down, no leaks
+
+
+def test_trainer_predict_over_launcher(tmp_path):
+    """trainer.predict with a remote strategy transports rank-0's
+    predictions back to the driver."""
+    model = BoringModel()
+    trainer = get_trainer(str(tmp_path), strategy=RayStrategy(num_workers=2),
+                          limit_predict_batches=2)
+    trainer.fit(model)
+    preds = trainer.predict(model)
+    assert isinstance(preds, list) and len(preds) == 2
+    assert preds[0].shape[-1] == 2
+
+
+def test_trainer_validate_over_launcher(tmp_path):
+    model = BoringModel()
+    trainer = get_trainer(str(tmp_path), strategy=RayStrategy(num_workers=2))
+    trainer.fit(model)
+    out = trainer.validate(model)
+    assert "x" in out[0]
+
+
+def test_precision_bf16_cpu(tmp_path):
+    model = BoringModel()
+    trainer = get_trainer(str(tmp_path), precision="bf16",
+                          checkpoint_callback=False)
+    trainer.fit(model)
+    assert trainer.state.finished
