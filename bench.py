@@ -58,7 +58,14 @@ def build_comm(rank: int, world: int, device: torch.device):
         from ray_lightning_amd.engine.rccl import (NativeRcclCommunicator,
                                                    rccl_available)
         if rccl_available():
-            return control, NativeRcclCommunicator(control, device)
+            try:
+                return control, NativeRcclCommunicator(control, device)
+            except Exception as e:  # noqa: BLE001 — never strand an
+                # 8-GPU run on an init quirk; torch-dist IS RCCL too
+                if rank == 0:
+                    print(f"[bench] native RCCL init failed ({e}); "
+                          "falling back to torch-dist RCCL",
+                          flush=True)
         import torch.distributed as dist
         return control, TorchDistCommunicator(
             dist.new_group(backend="nccl"))

@@ -202,10 +202,15 @@ class RayStrategy(Strategy):
         if self.use_gpu and whole_gpus and torch.cuda.is_available() \
                 and backend_override != "gloo":
             from ..engine.rccl import NativeRcclCommunicator, rccl_available
+            native_ok = False
             if rccl_available():
-                self._data_comm = NativeRcclCommunicator(
-                    self._comm, device=self.root_device)
-            else:
+                try:
+                    self._data_comm = NativeRcclCommunicator(
+                        self._comm, device=self.root_device)
+                    native_ok = True
+                except Exception:  # noqa: BLE001 — fall back below
+                    native_ok = False
+            if not native_ok:
                 # torch-dist RCCL fallback: still RCCL over xGMI, managed
                 # by torch instead of our extension.
                 import torch.distributed as dist
