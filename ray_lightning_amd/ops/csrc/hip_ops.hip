@@ -634,7 +634,7 @@ std::vector<torch::Tensor> fused_bn_fwd(
     c10::optional<torch::Tensor> res, bool relu, bool training,
     double momentum, double eps);
 std::vector<torch::Tensor> fused_bn_bwd(
-    torch::Tensor dy, torch::Tensor y, torch::Tensor x,
+    torch::Tensor dy, torch::Tensor mask, torch::Tensor x,
     torch::Tensor mean, torch::Tensor invstd, torch::Tensor gamma,
     bool relu, bool has_res);
 

@@ -40,21 +40,23 @@ class _FusedBNFunction(torch.autograd.Function):
                 running_var, training, momentum, eps, relu):
         ext = _load_ext()
         res_opt = residual
-        y, mean, invstd = ext.fused_bn_fwd(
+        y, mean, invstd, mask = ext.fused_bn_fwd(
             x, weight, bias,
             running_mean if running_mean is not None else torch.Tensor(),
             running_var if running_var is not None else torch.Tensor(),
             res_opt, relu, training, momentum, eps)
         ctx.relu = relu
         ctx.has_res = residual is not None
-        ctx.save_for_backward(x, y, mean, invstd, weight)
+        # the ReLU bitmask (1 byte / 8 elems) replaces saving+re-reading
+        # y in backward
+        ctx.save_for_backward(x, mask, mean, invstd, weight)
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, y, mean, invstd, weight = ctx.saved_tensors
+        x, mask, mean, invstd, weight = ctx.saved_tensors
         ext = _load_ext()
-        outs = ext.fused_bn_bwd(dy, y, x, mean, invstd, weight,
+        outs = ext.fused_bn_bwd(dy, mask, x, mean, invstd, weight,
                                 ctx.relu, ctx.has_res)
         dx, dgamma, dbeta = outs[0], outs[1], outs[2]
         dres = outs[3] if ctx.has_res else None
