@@ -144,10 +144,13 @@ class ShardedOptimizer(torch.optim.Optimizer):
         local = self.optim.state_dict()
         if self.world_size <= 1:
             return {"consolidated": local, "world_size": 1}
-        # Map local param indices to global indices.
+        # Map local param indices to global indices. NOTE: local
+        # indices follow the INNER optimizer's group order (the greedy
+        # shard order differs — using it mislabels every entry).
         gidx = {p: i for i, p in enumerate(self._all_params)}
-        owned = self._shards[self.rank]
-        local_to_global = [gidx[p] for p in owned]
+        inner_params = [p for g in self.optim.param_groups
+                        for p in g["params"]]
+        local_to_global = [gidx[p] for p in inner_params]
         shard_payload = {"state": local["state"],
                          "local_to_global": local_to_global}
         gathered = [None] * self.world_size
@@ -175,11 +178,13 @@ class ShardedOptimizer(torch.optim.Optimizer):
     def load_state_dict(self, state_dict: Dict) -> None:
         full = state_dict.get("consolidated", state_dict)
         gstate = full.get("state", {})
-        # Select this rank's shard out of the consolidated state.
+        # Select this rank's shard out of the consolidated state
+        # (local indices = inner-optimizer group order).
         gidx = {p: i for i, p in enumerate(self._all_params)}
-        owned = self._shards[self.rank]
+        inner_params = [p for g in self.optim.param_groups
+                        for p in g["params"]]
         local_state = {}
-        for li, p in enumerate(owned):
+        for li, p in enumerate(inner_params):
             gi = gidx[p]
             if gi in gstate:
                 local_state[li] = gstate[gi]

@@ -88,6 +88,26 @@ def _worker(rank: int, port: int, mode: str, out_q) -> None:
                 ddp.finalize_backward()
                 out = [p.grad.clone() for p in model.parameters()]
             out_q.put((rank, "ok", [t.numpy() for t in out]))
+        elif mode == "sharded_state":
+            from ray_lightning_amd.engine.sharded import (ShardedDDP,
+                                                          ShardedOptimizer)
+            opt = torch.optim.Adam(model.parameters(), lr=0.01)
+            oss = ShardedOptimizer(opt, comm, bucket_cap_mb=0.0001)
+            sddp = ShardedDDP(model, comm, oss, bucket_cap_mb=0.0001)
+            for _step in range(3):
+                model(_rank_batch(rank)).pow(2).mean().backward()
+                sddp.finalize_backward()
+                oss.step()
+                oss.zero_grad()
+            sd = oss.state_dict()  # collective consolidation
+            if rank == 0:
+                state = sd["consolidated"]["state"]
+                out = {gi: {k: (v.numpy() if torch.is_tensor(v) else v)
+                            for k, v in st.items()}
+                       for gi, st in state.items()}
+                out_q.put((rank, "ok", out))
+            else:
+                out_q.put((rank, "ok", None))
         elif mode == "sharded":
             from ray_lightning_amd.engine.sharded import (ShardedDDP,
                                                           ShardedOptimizer)
@@ -186,3 +206,26 @@ def test_sharded_params_match_plain_sgd():
         for got, e in zip(results[rank], expected):
             assert torch.allclose(torch.from_numpy(got), e, atol=1e-5), \
                 f"rank {rank} param mismatch"
+
+
+def test_sharded_optimizer_state_consolidation():
+    """Consolidated sharded-Adam state == single-process Adam state on
+    the averaged gradients (every param present, exp_avg equal)."""
+    results = _run_workers("sharded_state")
+    state = results[0]
+    model = _make_model()
+    opt = torch.optim.Adam(model.parameters(), lr=0.01)
+    for _step in range(3):
+        opt.zero_grad()
+        losses = [model(_rank_batch(r)).pow(2).mean()
+                  for r in range(WORLD)]
+        (sum(losses) / WORLD).backward()
+        opt.step()
+    n_params = len(list(model.parameters()))
+    assert set(state.keys()) == set(range(n_params))
+    for gi, p in enumerate(model.parameters()):
+        ref = opt.state[p]
+        got = state[gi]
+        assert got["step"] == 3 or float(got["step"]) == 3.0
+        assert torch.allclose(torch.from_numpy(got["exp_avg"]),
+                              ref["exp_avg"], atol=1e-5), f"param {gi}"
