@@ -65,3 +65,45 @@ def test_metrics_transported_from_gpu_worker(tmp_path):
     trainer.fit(model, datamodule=XORDataModule())
     assert float(trainer.callback_metrics["avg_val_loss"]) == \
         pytest.approx(0.3)
+
+
+def test_sharded_strategy_fractional_gpu(tmp_path):
+    """RayShardedStrategy with two workers sharing one device: OSS
+    partition + reduce-to-owner + param broadcast on real HIP tensors
+    (gloo data plane)."""
+    import warnings
+    from ray_lightning_amd import RayShardedStrategy
+    model = BoringModel()
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        strategy = RayShardedStrategy(num_workers=2, use_gpu=True,
+                                      resources_per_worker={"GPU": 0.5})
+    trainer = get_trainer(str(tmp_path), strategy=strategy,
+                          limit_train_batches=4, limit_val_batches=2)
+    train_test(trainer, model)
+
+
+def test_tune_trial_on_gpu(tmp_path):
+    """One Tune trial running a GPU worker end-to-end (reference
+    test_tune.py:95-116 downsized to 1 GPU)."""
+    from ray_lightning_amd import tune
+    from ray_lightning_amd.tune import (TuneReportCallback,
+                                        get_tune_resources)
+
+    def train_fn(config):
+        model = BoringModel()
+        trainer = Trainer(
+            default_root_dir=config["root"], max_epochs=1,
+            strategy=RayStrategy(num_workers=1, use_gpu=True),
+            callbacks=[TuneReportCallback({"loss": "x"},
+                                          on="validation_end")],
+            limit_train_batches=4, limit_val_batches=2,
+            num_sanity_val_steps=0, enable_checkpointing=False)
+        trainer.fit(model)
+
+    analysis = tune.run(
+        train_fn, config={"root": str(tmp_path)},
+        resources_per_trial=get_tune_resources(num_workers=1,
+                                               use_gpu=True),
+        local_dir=str(tmp_path / "tune"), metric="loss", mode="min")
+    assert analysis.trials[0].last_result["training_iteration"] == 1
