@@ -254,3 +254,38 @@ def test_precision_bf16_cpu(tmp_path):
                           checkpoint_callback=False)
     trainer.fit(model)
     assert trainer.state.finished
+
+
+def test_distributed_sampler_injection_unit():
+    """inject_distributed_sampler: replicas/rank wired, shuffle=True
+    only for train, loader settings preserved, IterableDataset and
+    pre-sampled loaders untouched (reference test_ddp.py:179-211)."""
+    from torch.utils.data import DataLoader, DistributedSampler
+    from ray_lightning_amd.trainer.data import inject_distributed_sampler
+    from utils import RandomDataset
+
+    dl = DataLoader(RandomDataset(32, 64), batch_size=4, num_workers=0,
+                    drop_last=True)
+    train = inject_distributed_sampler(dl, num_replicas=4, rank=2,
+                                       shuffle=True)
+    assert isinstance(train.sampler, DistributedSampler)
+    assert train.sampler.num_replicas == 4
+    assert train.sampler.rank == 2
+    assert train.sampler.shuffle is True
+    assert train.batch_size == 4 and train.drop_last is True
+
+    ev = inject_distributed_sampler(dl, num_replicas=4, rank=1,
+                                    shuffle=False)
+    assert ev.sampler.shuffle is False
+
+    # single replica: untouched
+    assert inject_distributed_sampler(dl, 1, 0, True) is dl
+    # already distributed: untouched
+    assert inject_distributed_sampler(train, 4, 2, True) is train
+
+    class _It(torch.utils.data.IterableDataset):
+        def __iter__(self):
+            return iter(range(8))
+
+    it_dl = DataLoader(_It(), batch_size=2)
+    assert inject_distributed_sampler(it_dl, 4, 0, True) is it_dl
