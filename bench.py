@@ -33,6 +33,14 @@ import time
 # fallback kernels on MI355X: 212 vs 6014 samples/s measured
 # (profiles/r01_resnet50_1gpu_fastfind.md). Override via env if needed.
 
+# Drop the naive reference convolutions from MIOpen's find candidate
+# list: at batch 768 each naive wrw timing costs ~0.7 s and the find
+# phase spends ~3 GPU-minutes on them (profiles/prof9) while never
+# winning a shape. The igemm/CK solvers are unaffected.
+os.environ.setdefault("MIOPEN_DEBUG_CONV_DIRECT_NAIVE_CONV_FWD", "0")
+os.environ.setdefault("MIOPEN_DEBUG_CONV_DIRECT_NAIVE_CONV_BWD", "0")
+os.environ.setdefault("MIOPEN_DEBUG_CONV_DIRECT_NAIVE_CONV_WRW", "0")
+
 import torch
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
