@@ -1,11 +1,12 @@
 """Time the flash kernels individually on the GPT-2-XL shape."""
 import math
+import os
 import sys
 import time
 
 import torch
 
-sys.path.insert(0, ".")
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from ray_lightning_amd import ops  # noqa: E402
 
 ext = ops._load_ext()

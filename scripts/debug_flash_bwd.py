@@ -1,10 +1,11 @@
 """Isolate flash-attention backward errors per tensor and tile size."""
 import math
+import os
 import sys
 
 import torch
 
-sys.path.insert(0, ".")
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from ray_lightning_amd import ops  # noqa: E402
 
 ext = ops._load_ext()
