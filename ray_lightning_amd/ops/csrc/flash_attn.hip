@@ -95,8 +95,10 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
   const int wrow0 = qm0 + wave * WM;   // wave's first query row
   const int col0 = lane & 15;
 
-  __shared__ __bf16 lds_vt[HS * BN];          // V^T [hs][kv]  8 KB
-  __shared__ __bf16 lds_p[4][WM * BN];        // per-wave P    16 KB
+  // V^T double-buffered: tile i writes buf[i&1] while others still
+  // read buf[(i-1)&1] -> ONE barrier per tile instead of two
+  __shared__ __bf16 lds_vt[2][HS * BN];       // V^T [hs][kv] 16 KB
+  __shared__ __bf16 lds_p[4][WM * BN];        // per-wave P   16 KB
 
   bf16x8 qf[2][2];  // [m-half][k-chunk]
   #pragma unroll
@@ -121,8 +123,9 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
 
   const float l2e = 1.4426950408889634f * scale;
   const int kv_end = qm0 + BM;
-  for (int kn0 = 0; kn0 < kv_end && kn0 < T; kn0 += BN) {
-    stage_transposed(lds_vt, v, kn0);
+  int buf = 0;
+  for (int kn0 = 0; kn0 < kv_end && kn0 < T; kn0 += BN, buf ^= 1) {
+    stage_transposed(lds_vt[buf], v, kn0);
     __syncthreads();
     // waves whose every query row is below this KV tile skip compute
     const bool active = kn0 <= wrow0 + WM - 1;
@@ -210,7 +213,7 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
           bf16x8 vb;
           *reinterpret_cast<int4*>(&vb) =
               *reinterpret_cast<const int4*>(
-                  lds_vt + (16 * n + col0) * BN + kk * 32 +
+                  lds_vt[buf] + (16 * n + col0) * BN + kk * 32 +
                   (lane >> 4) * 8);
           #pragma unroll
           for (int mh = 0; mh < 2; ++mh)
@@ -219,7 +222,7 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
         }
       }
     }
-    __syncthreads();
+    // no trailing barrier: next tile writes the OTHER V^T buffer
   }
 
   // --- epilogue -----------------------------------------------------
