@@ -289,3 +289,46 @@ def test_distributed_sampler_injection_unit():
 
     it_dl = DataLoader(_It(), batch_size=2)
     assert inject_distributed_sampler(it_dl, 4, 0, True) is it_dl
+
+
+def test_gradient_accumulation_two_workers(tmp_path):
+    """accumulate_grad_batches with the remote strategy: no_sync
+    suppresses comm on micro-batches; optimizer steps = batches/accum."""
+    model = BoringModel()
+    trainer = Trainer(default_root_dir=str(tmp_path), max_epochs=1,
+                      strategy=RayStrategy(num_workers=2),
+                      limit_train_batches=8, limit_val_batches=0,
+                      accumulate_grad_batches=4, num_sanity_val_steps=0,
+                      enable_checkpointing=False)
+    trainer.fit(model)
+    assert trainer.global_step == 2  # recovered from rank 0
+
+
+def test_three_consecutive_fits(tmp_path):
+    """Actors are created and torn down per launch; three fits on one
+    trainer must work (reference repeated-fit capability)."""
+    model = BoringModel()
+    trainer = get_trainer(str(tmp_path), strategy=RayStrategy(num_workers=2),
+                          limit_train_batches=3, limit_val_batches=1)
+    for i in range(3):
+        trainer.fit(model)
+        assert trainer.state.finished, f"fit #{i} failed"
+
+
+def test_tune_max_concurrent_trials(tmp_path):
+    from ray_lightning_amd import tune
+
+    def train_fn(config):
+        model = BoringModel()
+        t = Trainer(default_root_dir=config["root"], max_epochs=1,
+                    limit_train_batches=2, limit_val_batches=1,
+                    num_sanity_val_steps=0, enable_checkpointing=False)
+        t.fit(model)
+        tune.report(loss=1.0)
+
+    analysis = tune.run(
+        train_fn, config={"root": str(tmp_path)}, num_samples=3,
+        max_concurrent_trials=1, local_dir=str(tmp_path / "t"),
+        metric="loss", mode="min")
+    assert all(t.status == "TERMINATED" for t in analysis.trials)
+    assert len(analysis.trials) == 3
