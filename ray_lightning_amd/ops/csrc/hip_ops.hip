@@ -646,6 +646,8 @@ std::vector<torch::Tensor> flash_attn_bwd(
     torch::Tensor dout, torch::Tensor q, torch::Tensor k,
     torch::Tensor v, torch::Tensor o, torch::Tensor lse, double scale);
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
+torch::Tensor perm_probe(torch::Tensor M, torch::Tensor B,
+                         long variant);
 
 // fused residual-add + LayerNorm — defined in fused_ln.hip
 std::vector<torch::Tensor> fused_ln_fwd(
@@ -672,6 +674,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("flash_attn_bwd", &flash_attn_bwd,
         "causal flash attention backward");
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
+  m.def("perm_probe", &perm_probe,
+        "C-layout -> A-fragment bpermute redistribution probe");
   m.def("scale_inplace", &scale_inplace, "flat *= s");
   m.def("scale_cast", &scale_cast, "dst_f32 = src_bf16 * s");
   m.def("fused_sgd", &fused_sgd, "fused multi-tensor SGD(momentum)");

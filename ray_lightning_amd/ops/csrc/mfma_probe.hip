@@ -60,3 +60,122 @@ torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
                      C.data_ptr<float>());
   return C;
 }
+
+// ---------------------------------------------------------------------
+// perm_probe: verifies the in-register C-layout -> A-fragment
+// redistribution (ds_bpermute + v_perm) that the round-2 flash
+// redesign needs to eliminate the P LDS round-trip
+// (profiles/r01_flash_attn_notes.md).
+//
+// A wave holds M[16][64] f32 in MFMA C layout (lane l owns rows
+// (l>>4)*4+r, col l&15 + 16n). The probe packs row-pairs with
+// v_cvt_pk_bf16_f32 (compiler form), redistributes to the A-fragment
+// layout (row l&15, k (l>>4)*8+i) with 2 bpermutes + 1 byte-perm per
+// output dword, runs C2 = M x B through mfma_16x16x32, and the test
+// compares against torch.matmul.
+// ---------------------------------------------------------------------
+namespace {
+
+using bf16x8_p = __attribute__((ext_vector_type(8))) __bf16;
+using f32x4_p = __attribute__((ext_vector_type(4))) float;
+
+__device__ __forceinline__ unsigned pack_bf16_pair(float lo, float hi) {
+  __bf16 l = (__bf16)lo, h = (__bf16)hi;
+  unsigned short ul = __builtin_bit_cast(unsigned short, l);
+  unsigned short uh = __builtin_bit_cast(unsigned short, h);
+  return (unsigned)ul | ((unsigned)uh << 16);
+}
+
+__global__ void perm_probe_kernel(const float* __restrict__ M,
+                                  const __hip_bfloat16* __restrict__ B,
+                                  float* __restrict__ C2, int variant) {
+  const int l = threadIdx.x;
+  if (l >= 64) return;
+  const int g = l >> 4;        // lane group
+  const int c15 = l & 15;
+
+  // 1. load M into the C layout this lane would hold after an MFMA
+  float s[4][4];  // [r][n]: element (row=(g*4+r), col=c15+16n)
+  #pragma unroll
+  for (int r = 0; r < 4; ++r)
+    #pragma unroll
+    for (int n = 0; n < 4; ++n)
+      s[r][n] = M[(g * 4 + r) * 64 + c15 + 16 * n];
+
+  // 2. pack row-pairs: p[q][n] = bf16(row 4g+2q, col) | bf16(row
+  //    4g+2q+1, col) << 16
+  unsigned p[2][4];
+  #pragma unroll
+  for (int q = 0; q < 2; ++q)
+    #pragma unroll
+    for (int n = 0; n < 4; ++n)
+      p[q][n] = pack_bf16_pair(s[2 * q][n], s[2 * q + 1][n]);
+
+  // 3. redistribute to A-fragment layout: target lane l needs
+  //    A[row=c15][k=kk*32+g*8+i]; source of element (row, col):
+  //    lane (row>>2)*16 + (col&15), reg q=(row&3)>>1, n=col>>4,
+  //    half = row&1.
+  const int q_t = (c15 & 3) >> 1;          // fixed per target lane
+  const int parity = c15 & 1;
+  // perm byte pool: {second operand bytes 0-3, first operand bytes
+  // 4-7} per llvm.amdgcn.perm; variant 1 assumes the opposite order —
+  // the host test reports which matches (both computed in one pass,
+  // selected by the `variant` flag).
+  const unsigned sel0 = parity ? 0x07060302u : 0x05040100u;
+  const unsigned sel1 = parity ? 0x03020706u : 0x01000504u;
+  bf16x8_p afrag[2];
+  #pragma unroll
+  for (int kk = 0; kk < 2; ++kk) {
+    const int n_t = (kk * 32 + g * 8) >> 4;  // fixed per (lane, kk)
+    unsigned out[4];
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const int col_e = kk * 32 + g * 8 + 2 * j;
+      const int lane_e = (c15 >> 2) * 16 + (col_e & 15);
+      // ds_bpermute wants byte addresses (lane*4)
+      unsigned A = __builtin_amdgcn_ds_bpermute(lane_e * 4,
+                                                p[q_t][n_t]);
+      unsigned Bv = __builtin_amdgcn_ds_bpermute((lane_e + 1) * 4,
+                                                 p[q_t][n_t]);
+      out[j] = variant
+          ? __builtin_amdgcn_perm(Bv, A, sel1)
+          : __builtin_amdgcn_perm(Bv, A, sel0);
+    }
+    afrag[kk] = __builtin_bit_cast(bf16x8_p,
+        (__attribute__((ext_vector_type(4))) unsigned){
+            out[0], out[1], out[2], out[3]});
+  }
+
+  // 4. MFMA against B [64, 16] and write C2 = M x B
+  f32x4_p acc = {0.f, 0.f, 0.f, 0.f};
+  #pragma unroll
+  for (int kk = 0; kk < 2; ++kk) {
+    bf16x8_p bf;
+    #pragma unroll
+    for (int i = 0; i < 8; ++i)
+      bf[i] = *reinterpret_cast<const __bf16*>(
+          &B[(kk * 32 + g * 8 + i) * 16 + c15]);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[kk], bf, acc,
+                                                  0, 0, 0);
+  }
+  #pragma unroll
+  for (int r = 0; r < 4; ++r)
+    C2[(g * 4 + r) * 16 + c15] = acc[r];
+}
+
+}  // namespace
+
+torch::Tensor perm_probe(torch::Tensor M, torch::Tensor B,
+                         long variant) {
+  TORCH_CHECK(M.sizes() == torch::IntArrayRef({16, 64}) &&
+              M.scalar_type() == at::kFloat &&
+              B.sizes() == torch::IntArrayRef({64, 16}) &&
+              B.scalar_type() == at::kBFloat16);
+  auto C2 = torch::zeros({16, 16}, M.options());
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(perm_probe_kernel, dim3(1), dim3(64), 0, stream,
+                     M.data_ptr<float>(),
+                     reinterpret_cast<const __hip_bfloat16*>(B.data_ptr()),
+                     C2.data_ptr<float>(), (int)variant);
+  return C2;
+}

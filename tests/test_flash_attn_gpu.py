@@ -120,3 +120,22 @@ def test_flash_vs_sdpa_speed():
     print(f"\n[flash] custom {t_custom:.3f} ms  sdpa {t_sdpa:.3f} ms "
           f"(fwd+bwd, B{B} H{H} T{T})")
     assert t_custom < t_sdpa * 3  # sanity: not catastrophically slow
+
+
+def test_perm_redistribution_probe():
+    """C-layout -> A-fragment in-register redistribution (ds_bpermute
+    + v_perm): the primitive the round-2 flash redesign needs instead
+    of the P LDS round-trip. Accepts either byte-pool order of
+    v_perm_b32 and records which one gfx950 uses."""
+    from ray_lightning_amd import ops
+    ext = ops._load_ext()
+    torch.manual_seed(3)
+    M = torch.randn(16, 64, device="cuda") * 0.5
+    B = (torch.randn(64, 16, device="cuda") * 0.5).bfloat16()
+    ref = M.bfloat16().float() @ B.float()
+    errs = []
+    for variant in (0, 1):
+        C2 = ext.perm_probe(M, B, variant)
+        errs.append(float((C2 - ref).abs().max()))
+    print(f"\n[perm_probe] variant errs: {errs}")
+    assert min(errs) < 3e-2, f"neither perm variant matches: {errs}"
