@@ -648,6 +648,7 @@ std::vector<torch::Tensor> flash_attn_bwd(
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
 torch::Tensor perm_probe(torch::Tensor M, torch::Tensor B,
                          long variant);
+std::vector<torch::Tensor> perm_dump(torch::Tensor M, long variant);
 
 // fused residual-add + LayerNorm — defined in fused_ln.hip
 std::vector<torch::Tensor> fused_ln_fwd(
@@ -676,6 +677,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
   m.def("perm_probe", &perm_probe,
         "C-layout -> A-fragment bpermute redistribution probe");
+  m.def("perm_dump", &perm_dump, "redistribution element dump");
   m.def("scale_inplace", &scale_inplace, "flat *= s");
   m.def("scale_cast", &scale_cast, "dst_f32 = src_bf16 * s");
   m.def("fused_sgd", &fused_sgd, "fused multi-tensor SGD(momentum)");

@@ -179,3 +179,70 @@ torch::Tensor perm_probe(torch::Tensor M, torch::Tensor B,
                      C2.data_ptr<float>(), (int)variant);
   return C2;
 }
+
+// Diagnostic: dump the redistributed A-fragment (and a raw bpermute
+// identity check) so a failing perm_probe pinpoints the wrong stage.
+namespace {
+__global__ void perm_dump_kernel(const float* __restrict__ M,
+                                 float* __restrict__ out,
+                                 unsigned* __restrict__ ident,
+                                 int variant) {
+  const int l = threadIdx.x;
+  if (l >= 64) return;
+  const int g = l >> 4;
+  const int c15 = l & 15;
+  // bpermute identity: lane l fetches lane (l+1)%64's value l+100
+  ident[l] = __builtin_amdgcn_ds_bpermute(((l + 1) % 64) * 4,
+                                          (unsigned)(l + 100));
+  float s[4][4];
+  #pragma unroll
+  for (int r = 0; r < 4; ++r)
+    #pragma unroll
+    for (int n = 0; n < 4; ++n)
+      s[r][n] = M[(g * 4 + r) * 64 + c15 + 16 * n];
+  unsigned p[2][4];
+  #pragma unroll
+  for (int q = 0; q < 2; ++q)
+    #pragma unroll
+    for (int n = 0; n < 4; ++n)
+      p[q][n] = pack_bf16_pair(s[2 * q][n], s[2 * q + 1][n]);
+  const int q_t = (c15 & 3) >> 1;
+  const int parity = c15 & 1;
+  const unsigned sel0 = parity ? 0x07060302u : 0x05040100u;
+  const unsigned sel1 = parity ? 0x03020706u : 0x01000504u;
+  #pragma unroll
+  for (int kk = 0; kk < 2; ++kk) {
+    const int n_t = (kk * 32 + g * 8) >> 4;
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const int col_e = kk * 32 + g * 8 + 2 * j;
+      const int lane_e = (c15 >> 2) * 16 + (col_e & 15);
+      unsigned A = __builtin_amdgcn_ds_bpermute(lane_e * 4,
+                                                p[q_t][n_t]);
+      unsigned Bv = __builtin_amdgcn_ds_bpermute((lane_e + 1) * 4,
+                                                 p[q_t][n_t]);
+      unsigned o = variant ? __builtin_amdgcn_perm(Bv, A, sel1)
+                           : __builtin_amdgcn_perm(Bv, A, sel0);
+      __bf16 lo = __builtin_bit_cast(__bf16,
+          (unsigned short)(o & 0xffff));
+      __bf16 hi = __builtin_bit_cast(__bf16,
+          (unsigned short)(o >> 16));
+      // element (row=c15, k=col_e) and (row=c15, k=col_e+1)
+      out[(long)c15 * 64 + col_e] = (float)lo;
+      out[(long)c15 * 64 + col_e + 1] = (float)hi;
+    }
+  }
+}
+}  // namespace
+
+std::vector<torch::Tensor> perm_dump(torch::Tensor M, long variant) {
+  auto out = torch::zeros({16, 64}, M.options());
+  auto ident = torch::zeros({64},
+                            M.options().dtype(at::kInt));
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(perm_dump_kernel, dim3(1), dim3(64), 0, stream,
+                     M.data_ptr<float>(), out.data_ptr<float>(),
+                     reinterpret_cast<unsigned*>(ident.data_ptr()),
+                     (int)variant);
+  return {out, ident};
+}
