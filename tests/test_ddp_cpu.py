@@ -332,3 +332,22 @@ def test_tune_max_concurrent_trials(tmp_path):
         metric="loss", mode="min")
     assert all(t.status == "TERMINATED" for t in analysis.trials)
     assert len(analysis.trials) == 3
+
+
+def test_two_worker_training_is_deterministic(tmp_path):
+    """Same seed, same config => bitwise-identical weights across two
+    independent 2-worker runs (the engines use fixed-order reductions;
+    no atomics in the CPU path)."""
+    results = []
+    for run in range(2):
+        import ray_lightning_amd as rla
+        rla.seed_everything(123)
+        model = BoringModel()
+        trainer = get_trainer(str(tmp_path / f"run{run}"),
+                              strategy=RayStrategy(num_workers=2),
+                              limit_train_batches=4, limit_val_batches=1,
+                              checkpoint_callback=False)
+        trainer.fit(model)
+        results.append(torch.cat(
+            [p.detach().flatten() for p in model.parameters()]))
+    assert torch.equal(results[0], results[1])
