@@ -8,6 +8,12 @@ pytestmark = pytest.mark.gpu
 if not torch.cuda.is_available():
     pytest.skip("needs MI355X", allow_module_level=True)
 
+# the fused path must be the one under test — fail loudly if the HIP
+# extension did not load on a GPU box (no silent eager fallback)
+from ray_lightning_amd import ops as _ops
+assert _ops._load_ext() is not None, "HIP extension failed to load"
+
+
 
 def _ref(x32, w, b, rm, rv, res32, relu, training, momentum, eps):
     """fp32 reference: torch BN + add + relu (keeps grads)."""

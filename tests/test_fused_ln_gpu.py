@@ -8,6 +8,12 @@ pytestmark = pytest.mark.gpu
 if not torch.cuda.is_available():
     pytest.skip("needs MI355X", allow_module_level=True)
 
+# the fused path must be the one under test — fail loudly if the HIP
+# extension did not load on a GPU box (no silent eager fallback)
+from ray_lightning_amd import ops as _ops
+assert _ops._load_ext() is not None, "HIP extension failed to load"
+
+
 
 @pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float32])
 @pytest.mark.parametrize("C", [1600, 768, 2048])
