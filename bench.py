@@ -172,8 +172,6 @@ def main() -> None:
         def fwd(i: int):
             _, loss = model(xs[i % n_buf], ys[i % n_buf])
             return loss
-        autocast_needed = False
-        channels_last = False
     else:
         model = resnet50(args.num_classes).to(device)
         channels_last = args.memory_format == "channels_last" and on_gpu
@@ -211,7 +209,6 @@ def main() -> None:
                 logits = model(images[i % n_buf])
                 return torch.nn.functional.cross_entropy(
                     logits.float(), labels[i % n_buf])
-        autocast_needed = True
 
     def step(i: int) -> None:
         loss = fwd(i)

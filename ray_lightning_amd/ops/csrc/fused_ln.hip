@@ -85,7 +85,6 @@ __global__ __launch_bounds__(kT) void ln_fwd_kernel(
     // pass 1: x = a (+ b); accumulate sum/sumsq; stage x in registers
     // when the row fits (C <= kT*V covers GPT-2 sizes), else re-read.
     float s = 0.f, s2 = 0.f;
-    T xreg[LnVec<T>::V];  // per-iteration staging
     const bool fits = nvec <= kT;
     // local storage for one vector slot per thread (fits case)
     T xloc[LnVec<T>::V];
@@ -114,7 +113,6 @@ __global__ __launch_bounds__(kT) void ln_fwd_kernel(
         for (int i = 0; i < V; ++i) xloc[i] = av[i];
       }
     }
-    (void)xreg;
     s = block_sum(s, lds);
     s2 = block_sum(s2, lds);
     const float mean = s / (float)C;
