@@ -351,3 +351,28 @@ def test_two_worker_training_is_deterministic(tmp_path):
         results.append(torch.cat(
             [p.detach().flatten() for p in model.parameters()]))
     assert torch.equal(results[0], results[1])
+
+
+def test_executable_cls_override_seam():
+    """launchers.utils.get_executable_cls test seam (reference
+    launchers/utils.py:20-24): a custom actor class is used for
+    workers when set."""
+    from ray_lightning_amd.launchers import utils as lutils
+    from ray_lightning_amd.runtime import ActorHandle
+
+    created = []
+
+    class _SpyActor(ActorHandle):
+        def __init__(self, env, name="spy"):
+            created.append(name)
+            super().__init__(env, name=name)
+
+    lutils.set_executable_cls(_SpyActor)
+    try:
+        strategy = RayStrategy(num_workers=2)
+        launcher = RayLauncher(strategy)
+        launcher.setup_workers()
+        launcher.teardown_workers()
+    finally:
+        lutils.set_executable_cls(None)
+    assert len(created) == 2

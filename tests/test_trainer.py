@@ -147,3 +147,25 @@ def test_checkpoint_contains_trainer_state(tmp_path):
     assert "state_dict" in ckpt
     assert "optimizer_states" in ckpt
     assert ckpt["epoch"] == 0  # saved during epoch 0's val end
+
+
+def test_check_val_every_n_epoch(tmp_path):
+    model = XORModel()
+    dm = XORDataModule()
+    counted = {"val_epochs": 0}
+
+    class _CountVal(BoringModel):
+        pass
+
+    from ray_lightning_amd import Callback
+
+    class _Counter(Callback):
+        def on_validation_epoch_end(self, trainer, pl_module):
+            if not trainer.sanity_checking:
+                counted["val_epochs"] += 1
+
+    trainer = Trainer(default_root_dir=str(tmp_path), max_epochs=4,
+                      check_val_every_n_epoch=2, callbacks=[_Counter()],
+                      enable_checkpointing=False, num_sanity_val_steps=0)
+    trainer.fit(model, datamodule=dm)
+    assert counted["val_epochs"] == 2  # epochs 2 and 4
