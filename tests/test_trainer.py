@@ -169,3 +169,12 @@ def test_check_val_every_n_epoch(tmp_path):
                       enable_checkpointing=False, num_sanity_val_steps=0)
     trainer.fit(model, datamodule=dm)
     assert counted["val_epochs"] == 2  # epochs 2 and 4
+
+
+def test_limit_train_batches_fraction(tmp_path):
+    model = BoringModel()  # 16 batches of 4 in its loader
+    trainer = Trainer(default_root_dir=str(tmp_path), max_epochs=1,
+                      limit_train_batches=0.25, limit_val_batches=0,
+                      num_sanity_val_steps=0, enable_checkpointing=False)
+    trainer.fit(model)
+    assert trainer.global_step == 4  # 16 * 0.25

@@ -96,3 +96,19 @@ def test_search_space_sampling(tmp_path):
         {"x": tune.grid_search([1, 2]), "y": tune.grid_search([3, 4])},
         num_samples=1, seed=0)
     assert len(grid) == 4
+
+
+def test_report_outside_session_raises():
+    with pytest.raises(RuntimeError, match="outside a Tune session"):
+        tune.report(loss=1.0)
+
+
+def test_checkpoint_dir_outside_session_raises():
+    with pytest.raises(RuntimeError, match="outside a Tune session"):
+        with tune.checkpoint_dir(step=0):
+            pass
+
+
+def test_tune_callback_invalid_on_raises():
+    with pytest.raises(ValueError, match="Invalid `on`"):
+        TuneReportCallback(on="nonsense_hook")
