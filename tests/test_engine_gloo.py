@@ -88,6 +88,24 @@ def _worker(rank: int, port: int, mode: str, out_q) -> None:
                 ddp.finalize_backward()
                 out = [p.grad.clone() for p in model.parameters()]
             out_q.put((rank, "ok", [t.numpy() for t in out]))
+        elif mode == "buffers":
+            from ray_lightning_amd.engine.ddp import NativeDDP
+            bn_model = torch.nn.Sequential(
+                torch.nn.Linear(16, 8), torch.nn.BatchNorm1d(8),
+                torch.nn.Linear(8, 2))
+            torch.manual_seed(rank)  # different init AND buffers
+            for prm in bn_model.parameters():
+                with torch.no_grad():
+                    prm.add_(torch.randn_like(prm) * 0.1)
+            ddp = NativeDDP(bn_model, comm, bucket_cap_mb=0.0001)
+            bn_model.train()
+            bn_model(_rank_batch(rank)).pow(2).mean().backward()
+            ddp.finalize_backward()
+            bn = bn_model[1]
+            out_q.put((rank, "ok", {
+                "running_mean": bn.running_mean.numpy(),
+                "weight": bn_model[0].weight.detach().numpy(),
+                "grad0": bn_model[0].weight.grad.numpy()}))
         elif mode == "sharded_state":
             from ray_lightning_amd.engine.sharded import (ShardedDDP,
                                                           ShardedOptimizer)
@@ -229,3 +247,16 @@ def test_sharded_optimizer_state_consolidation():
         assert got["step"] == 3 or float(got["step"]) == 3.0
         assert torch.allclose(torch.from_numpy(got["exp_avg"]),
                               ref["exp_avg"], atol=1e-5), f"param {gi}"
+
+
+def test_ddp_buffer_broadcast_and_grads():
+    """Wrap-time broadcast equalizes params AND buffers from rank 0;
+    per-rank BN stats then diverge locally (torch DDP semantics) while
+    grads are averaged identically."""
+    results = _run_workers("buffers")
+    w0 = torch.from_numpy(results[0]["weight"])
+    w1 = torch.from_numpy(results[1]["weight"])
+    assert torch.equal(w0, w1)  # replicas equalized at wrap time
+    g0 = torch.from_numpy(results[0]["grad0"])
+    g1 = torch.from_numpy(results[1]["grad0"])
+    assert torch.allclose(g0, g1, atol=1e-6)  # averaged grads agree
