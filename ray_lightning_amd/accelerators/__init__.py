@@ -1,12 +1,14 @@
-"""Delayed-GPU accelerator (reference accelerators/ parity).
+"""Accelerator registry (reference accelerators/ parity, consumed here).
 
 The reference registers a ``_GPUAccelerator`` under the name ``"_gpu"``
 whose ``is_available`` is always True and whose device binding is
 deferred until code runs inside a worker (reference
 accelerators/delayed_gpu_accelerator.py:47-60): the *driver* may be a
-CPU-only laptop while the workers have GPUs. In this framework the same
-semantics live in the strategies (``RayStrategy.root_device`` resolves
-worker-side only); these classes keep the public registry surface.
+CPU-only laptop while the workers have GPUs. Strategies publish their
+accelerator name (``Strategy.accelerator``) and the Trainer resolves
+device setup through ``ACCELERATOR_REGISTRY`` at stage start — the same
+resolution shape as PTL's accelerator registry (reference
+accelerators/__init__.py:13-19, ray_ddp.py:112-113).
 """
 from __future__ import annotations
 
@@ -15,11 +17,10 @@ import torch
 ACCELERATOR_REGISTRY = {}
 
 
-class _GPUAccelerator:
-    """Registered as "_gpu": availability is asserted on the driver even
-    without a local GPU; device binding happens in the worker."""
+class CPUAccelerator:
+    """Default accelerator: no device binding needed."""
 
-    name = "_gpu"
+    name = "cpu"
 
     @staticmethod
     def is_available() -> bool:
@@ -28,12 +29,27 @@ class _GPUAccelerator:
     @staticmethod
     def setup_device(device: torch.device) -> None:
         if device.type == "cuda":
+            # a CPU-strategy trainer may still land on a worker whose
+            # root_device resolved to a GPU (use_gpu after the fact) —
+            # bind it rather than silently computing on device 0
             torch.cuda.set_device(device)
 
     @classmethod
     def register_accelerators(cls, registry=None) -> None:
         (registry if registry is not None
          else ACCELERATOR_REGISTRY)[cls.name] = cls
+
+
+class _GPUAccelerator(CPUAccelerator):
+    """Registered as "_gpu": availability is asserted on the driver even
+    without a local GPU; device binding happens in the worker."""
+
+    name = "_gpu"
+
+    @staticmethod
+    def setup_device(device: torch.device) -> None:
+        if device.type == "cuda":
+            torch.cuda.set_device(device)
 
 
 class DelayedGPUAccelerator(_GPUAccelerator):
@@ -49,7 +65,18 @@ class DelayedGPUAccelerator(_GPUAccelerator):
         torch.cuda.set_device(device)
 
 
+def resolve_accelerator(name: str):
+    """Look up an accelerator by registry name (Trainer entry point)."""
+    try:
+        return ACCELERATOR_REGISTRY[name]
+    except KeyError:
+        raise KeyError(
+            f"Unknown accelerator {name!r}; registered: "
+            f"{sorted(ACCELERATOR_REGISTRY)}") from None
+
+
+CPUAccelerator.register_accelerators()
 _GPUAccelerator.register_accelerators()
 
-__all__ = ["_GPUAccelerator", "DelayedGPUAccelerator",
-           "ACCELERATOR_REGISTRY"]
+__all__ = ["CPUAccelerator", "_GPUAccelerator", "DelayedGPUAccelerator",
+           "ACCELERATOR_REGISTRY", "resolve_accelerator"]

@@ -52,6 +52,13 @@ class ShardedOptimizer(torch.optim.Optimizer):
         self.rank = comm.rank
         self._all_params: List[nn.Parameter] = [
             p for g in optimizer.param_groups for p in g["params"]]
+        # Original group membership as global indices — consolidated
+        # checkpoints must keep the param→group assignment (per-group
+        # lr/weight_decay) like torch/fairscale, not claim every index in
+        # every group.
+        _gidx = {p: i for i, p in enumerate(self._all_params)}
+        self._group_global_indices: List[List[int]] = [
+            [_gidx[p] for p in g["params"]] for g in optimizer.param_groups]
         self._shards = partition_params(self._all_params, self.world_size)
         self._owner: Dict[nn.Parameter, int] = {}
         for r, shard in enumerate(self._shards):
@@ -170,9 +177,10 @@ class ShardedOptimizer(torch.optim.Optimizer):
                     state[gi] = payload["state"][str(li)]
         return {"consolidated": {"state": state,
                                  "param_groups": [
-                                     {**g, "params": list(range(
-                                         len(self._all_params)))}
-                                     for g in self.optim.param_groups]},
+                                     {**g, "params": list(gi)}
+                                     for g, gi in zip(
+                                         self.optim.param_groups,
+                                         self._group_global_indices)]},
                 "world_size": self.world_size}
 
     def load_state_dict(self, state_dict: Dict) -> None:
@@ -194,8 +202,13 @@ class ShardedOptimizer(torch.optim.Optimizer):
         inner["state"] = local_state
         try:
             self.optim.load_state_dict(inner)
-        except (ValueError, KeyError):
-            pass
+        except (ValueError, KeyError) as e:
+            from ..util import rank_zero_warn
+            rank_zero_warn(
+                f"Sharded optimizer state restore dropped for "
+                f"{type(self.optim).__name__} ({type(e).__name__}: {e}); "
+                "continuing with fresh state (expected only on "
+                "world-size-change resume).")
 
 
 class ShardedDDP(nn.Module):

@@ -1,7 +1,8 @@
 """Hand-written CDNA4 MFMA flash attention (causal, hs=64, bf16).
 
 ``flash_attention(q, k, v)`` is SDPA-shaped: [B, H, T, hs] in/out.
-The custom kernels engage for causal bf16 hs=64 T%64==0 on GPU;
+The custom kernels engage for causal bf16 hs=64 T%128==0 on GPU
+(BM=128 row tiles);
 anything else falls back to ``F.scaled_dot_product_attention``.
 ``RLA_FLASH=0`` disables the custom path globally.
 """

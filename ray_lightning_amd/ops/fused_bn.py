@@ -91,9 +91,19 @@ class FusedBatchNorm2d(nn.BatchNorm2d):
             if training and self.track_running_stats and \
                     self.num_batches_tracked is not None:
                 self.num_batches_tracked.add_(1)
+            # momentum=None means cumulative moving average
+            # (nn.BatchNorm2d semantics): effective factor is
+            # 1/num_batches_tracked — the kernel binds a double.
+            momentum = self.momentum
+            if momentum is None:
+                momentum = (1.0 / float(self.num_batches_tracked)
+                            if (self.track_running_stats and
+                                self.num_batches_tracked is not None and
+                                int(self.num_batches_tracked) > 0)
+                            else 0.0)
             return _FusedBNFunction.apply(
                 x, residual, self.weight, self.bias, self.running_mean,
-                self.running_var, training, self.momentum, self.eps,
+                self.running_var, training, momentum, self.eps,
                 self.relu)
         # composed fallback (CPU, odd channel counts, missing ext)
         out = super().forward(x)

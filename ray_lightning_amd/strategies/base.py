@@ -57,6 +57,12 @@ class Strategy:
         return torch.device("cpu")
 
     @property
+    def accelerator(self) -> str:
+        """Registry name the Trainer resolves device setup through
+        (reference ray_ddp.py:112-113 selects "_gpu"/"cpu")."""
+        return "cpu"
+
+    @property
     def distributed_sampler_kwargs(self) -> Optional[Dict[str, int]]:
         return None
 

@@ -92,6 +92,12 @@ class RayStrategy(Strategy):
     def is_remote_launch(self) -> bool:  # type: ignore[override]
         return not self._external_mode
 
+    @property
+    def accelerator(self) -> str:
+        """"_gpu" defers device binding to the worker (the driver may be
+        CPU-only) — reference ray_ddp.py:112-113."""
+        return "_gpu" if self.use_gpu else "cpu"
+
     def set_remote(self, remote: bool) -> None:
         self._is_remote = remote
 

@@ -69,10 +69,16 @@ class CheckpointConnector:
                               ckpt.get("optimizer_states", [])):
             try:
                 opt.load_state_dict(state)
-            except (ValueError, KeyError):
+            except (ValueError, KeyError) as e:
                 # Param-group mismatch (e.g. resume with a different
-                # worker count onto a sharded optimizer): keep fresh state.
-                pass
+                # worker count onto a sharded optimizer): keep fresh state
+                # — but say so, so corruption is distinguishable from an
+                # intentional world-size change.
+                from ..util import rank_zero_warn
+                rank_zero_warn(
+                    f"Dropped optimizer state for {type(opt).__name__} "
+                    f"on resume ({type(e).__name__}: {e}); continuing "
+                    "with fresh optimizer state.")
         for sched, state in zip(trainer.lr_schedulers,
                                 ckpt.get("lr_schedulers", [])):
             sched.load_state_dict(state)

@@ -311,11 +311,15 @@ class Trainer:
         model = self.model
         stage_name = self.state.fn.value if self.state.fn else "fit"
 
-        # 1. environment + device
+        # 1. environment + device — resolved through the accelerator
+        # registry (reference accelerators/__init__.py:13-19: the strategy
+        # names "_gpu"/"cpu", the registry owns the binding semantics)
         self.strategy.setup_environment()
         device = self.strategy.root_device
-        if device.type == "cuda":
-            torch.cuda.set_device(device)
+        from ..accelerators import resolve_accelerator
+        accelerator = resolve_accelerator(
+            getattr(self.strategy, "accelerator", "cpu"))
+        accelerator.setup_device(device)
 
         # 2. data preparation hooks (prepare_data is driver/once semantics —
         #    the launcher calls it pre-fanout; here we call setup()).

@@ -275,6 +275,12 @@ def run(train_fn: Callable,
     if need_gpus > 0 and total_gpus == 0:
         raise RuntimeError(
             f"Trials require {need_gpus} GPUs but none are visible.")
+    if need_gpus > total_gpus:
+        # a single trial can never fit — fail fast instead of spinning
+        # with every trial pending forever
+        raise RuntimeError(
+            f"Trials require {need_gpus} GPUs each but only "
+            f"{total_gpus} are visible on this node.")
 
     report_queue = Queue()
     pending = list(trials)
@@ -285,8 +291,14 @@ def run(train_fn: Callable,
     def _can_start() -> bool:
         if max_concurrent_trials and len(running) >= max_concurrent_trials:
             return False
+        # "or not running" escape: an overcommitted demand still makes
+        # progress one trial at a time (same escape for CPUs and GPUs —
+        # a stricter GPU gate deadlocked when demand == capacity left
+        # fractional dust)
         cpu_ok = used_cpus + need_cpus <= total_cpus or not running
-        gpu_ok = used_gpus + need_gpus <= total_gpus or need_gpus == 0
+        gpu_ok = (need_gpus == 0
+                  or used_gpus + need_gpus <= total_gpus
+                  or not running)
         return cpu_ok and gpu_ok
 
     while pending or running:

@@ -69,3 +69,42 @@ def test_gpu_accelerator_registry():
     assert ACCELERATOR_REGISTRY["_gpu"] is _GPUAccelerator
     # driver-side availability is asserted even without a local GPU
     assert _GPUAccelerator.is_available()
+
+
+def test_strategy_publishes_accelerator_name():
+    """Strategies name their accelerator (reference ray_ddp.py:112-113)
+    and the name resolves through the registry."""
+    from ray_lightning_amd.accelerators import resolve_accelerator
+    assert RayStrategy(num_workers=1, use_gpu=True).accelerator == "_gpu"
+    assert RayStrategy(num_workers=1, use_gpu=False).accelerator == "cpu"
+    for name in ("_gpu", "cpu"):
+        assert resolve_accelerator(name).is_available()
+    with pytest.raises(KeyError):
+        resolve_accelerator("tpu")
+
+
+def test_trainer_resolves_device_through_registry(monkeypatch):
+    """The Trainer's device binding goes through ACCELERATOR_REGISTRY —
+    a strategy with an unknown accelerator name fails loudly."""
+    from ray_lightning_amd.trainer import Trainer
+    from ray_lightning_amd.strategies.base import SingleDeviceStrategy
+    from utils import BoringModel
+
+    calls = []
+
+    class Spy:
+        name = "cpu"
+
+        @staticmethod
+        def is_available():
+            return True
+
+        @staticmethod
+        def setup_device(device):
+            calls.append(device)
+
+    monkeypatch.setitem(ACCELERATOR_REGISTRY, "cpu", Spy)
+    t = Trainer(fast_dev_run=1,
+                strategy=SingleDeviceStrategy(torch.device("cpu")))
+    t.fit(BoringModel())
+    assert calls == [torch.device("cpu")]

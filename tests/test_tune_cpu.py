@@ -112,3 +112,32 @@ def test_checkpoint_dir_outside_session_raises():
 def test_tune_callback_invalid_on_raises():
     with pytest.raises(ValueError, match="Invalid `on`"):
         TuneReportCallback(on="nonsense_hook")
+
+
+def test_tune_gpu_overcommit_raises_fast(tmp_path, monkeypatch):
+    """A per-trial GPU demand that exceeds node capacity must raise
+    immediately instead of spinning with every trial pending
+    (r01 deadlock: need_gpus=2 on a 1-GPU node hung forever)."""
+    monkeypatch.setenv("HIP_VISIBLE_DEVICES", "0")
+    with pytest.raises(RuntimeError, match="only 1.0 are visible"):
+        tune.run(
+            _train_fn,
+            name="overcommit",
+            config={"max_epochs": 1, "root": str(tmp_path)},
+            num_samples=2,
+            local_dir=str(tmp_path),
+            resources_per_trial=get_tune_resources(
+                num_workers=2, use_gpu=True))
+
+
+def test_tune_gpu_zero_visible_raises(tmp_path, monkeypatch):
+    monkeypatch.setenv("HIP_VISIBLE_DEVICES", "")
+    with pytest.raises(RuntimeError, match="none are visible"):
+        tune.run(
+            _train_fn,
+            name="nogpu",
+            config={"max_epochs": 1, "root": str(tmp_path)},
+            num_samples=1,
+            local_dir=str(tmp_path),
+            resources_per_trial=get_tune_resources(
+                num_workers=1, use_gpu=True))
