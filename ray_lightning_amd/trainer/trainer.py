@@ -607,7 +607,9 @@ class Trainer:
                 self._global_step += 1
                 self._step_schedulers(interval="step")
 
-            if isinstance(loss, torch.Tensor):
+            # float() of a CUDA tensor is a device sync every batch — only
+            # pay it when a progress bar actually displays the value
+            if self.enable_progress_bar and isinstance(loss, torch.Tensor):
                 self._logger_connector.progress_bar_metrics["loss"] = \
                     float(loss.detach())
             model.on_train_batch_end(out, batch, batch_idx)
