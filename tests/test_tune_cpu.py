@@ -163,8 +163,6 @@ else:
 
 # core training is unaffected (exercises the launcher's
 # is_session_enabled hook in degraded mode)
-import sys
-sys.path.insert(0, %r)
 from utils import BoringModel
 from ray_lightning_amd import RayStrategy, Trainer
 trainer = Trainer(max_epochs=1, limit_train_batches=2,
@@ -174,8 +172,13 @@ trainer = Trainer(max_epochs=1, limit_train_batches=2,
 trainer.fit(BoringModel())
 assert trainer.state.finished
 print("DEGRADED-OK")
-""" % os.path.dirname(os.path.abspath(__file__))
-    env = dict(os.environ, RLA_DISABLE_TUNE="1")
+"""
+    tests_dir = os.path.dirname(os.path.abspath(__file__))
+    repo_root = os.path.dirname(tests_dir)
+    env = dict(os.environ, RLA_DISABLE_TUNE="1",
+               PYTHONPATH=os.pathsep.join(
+                   [repo_root, tests_dir,
+                    os.environ.get("PYTHONPATH", "")]))
     out = subprocess.run([sys.executable, "-c", script], env=env,
                          capture_output=True, text=True, timeout=180,
                          cwd=str(tmp_path))
