@@ -1,0 +1,132 @@
+"""Remote-driver ("Ray Client"-shaped) topology tests.
+
+The reference runs its examples through a Ray Client connection: the
+user process holds only a thin client, the DRIVER code executes in a
+server-side process, and that driver spawns the worker pool (reference
+tests/test_client.py:1-31, test_client_2.py, test_client_3.py;
+README.md:105 caveats). Ray Client itself cannot exist here, but the
+capability under test — driver process != user process != workers, with
+the collect/recover protocol crossing those boundaries — is exercised
+with the framework's own actor runtime as the remote-driver host:
+the test process ("client") executes the train function inside a
+spawned driver actor, and THAT driver spawns the training workers.
+"""
+import os
+
+import pytest
+import torch
+
+from ray_lightning_amd.runtime import ActorHandle
+
+from utils import BoringModel
+
+
+def _remote_driver_train(strategy_name: str, num_workers: int,
+                         root: str) -> dict:
+    """Runs inside the driver-host process (the 'server side' of the
+    client topology). Builds the Trainer, fits through the actor
+    launcher, and returns what a client needs to verify recover."""
+    from ray_lightning_amd import (RayShardedStrategy, RayStrategy,
+                                   Trainer)
+    from ray_lightning_amd.strategies.ray_horovod import \
+        HorovodRayStrategy
+
+    cls = {"ddp": RayStrategy, "sharded": RayShardedStrategy,
+           "horovod": HorovodRayStrategy}[strategy_name]
+    model = BoringModel()
+    trainer = Trainer(
+        max_epochs=1, limit_train_batches=4, limit_val_batches=2,
+        num_sanity_val_steps=0, default_root_dir=root,
+        strategy=cls(num_workers=num_workers))
+    trainer.fit(model)
+    # driver-side recover already ran: weights + metrics + ckpt path
+    ckpt = trainer.checkpoint_callback
+    return {
+        "finished": trainer.state.finished,
+        "global_step": trainer.global_step,
+        "epoch": trainer.current_epoch,
+        "has_loss": "x" in trainer.callback_metrics
+                    or "val_loss" in trainer.callback_metrics
+                    or len(trainer.callback_metrics) > 0,
+        "best_model_path": ckpt.best_model_path if ckpt else None,
+        "weight_sum": float(
+            sum(p.abs().sum() for p in model.parameters())),
+    }
+
+
+@pytest.mark.parametrize("strategy_name", ["ddp", "sharded", "horovod"])
+def test_remote_driver_topology(tmp_path, strategy_name):
+    """Driver in its own process, workers in grandchild processes; the
+    client only sees the returned summary (reference test_client*.py
+    matrix: one test per strategy)."""
+    driver = ActorHandle({}, name="client-driver")
+    try:
+        fut = driver.execute(_remote_driver_train, strategy_name, 2,
+                             str(tmp_path))
+        out = fut.get(timeout=300)
+    finally:
+        driver.kill()
+    assert out["finished"]
+    assert out["global_step"] >= 1
+    assert out["epoch"] == 1
+    assert out["weight_sum"] > 0
+    if out["best_model_path"]:
+        # checkpoint written by the remote driver must be reachable from
+        # the client process (shared filesystem, like Ray's caveat that
+        # ckpt paths are server-side paths)
+        assert os.path.exists(out["best_model_path"])
+
+
+def test_remote_driver_exception_propagates(tmp_path):
+    """A failure inside the remote driver surfaces to the client with
+    the remote traceback (fail-fast across the client boundary)."""
+    from ray_lightning_amd.runtime.actor import RemoteError
+
+    driver = ActorHandle({}, name="client-driver-err")
+    try:
+        fut = driver.execute(_remote_driver_train, "nonsense", 1,
+                             str(tmp_path))
+        with pytest.raises((RemoteError, KeyError)):
+            fut.get(timeout=120)
+    finally:
+        driver.kill()
+
+
+def test_remote_driver_tune(tmp_path):
+    """Tune experiment driven from a remote driver process (reference
+    test_client.py:26-31 tune-through-client)."""
+    driver = ActorHandle({}, name="client-tune-driver")
+    try:
+        fut = driver.execute(_remote_tune, str(tmp_path))
+        out = fut.get(timeout=300)
+    finally:
+        driver.kill()
+    assert out["n_trials"] == 2
+    assert all(it == 1 for it in out["iters"])
+
+
+def _remote_tune(root: str) -> dict:
+    from ray_lightning_amd import tune
+
+    analysis = tune.run(
+        _tune_train_fn,
+        config={"root": root, "max_epochs": tune.grid_search([1, 1])},
+        local_dir=os.path.join(root, "tune"),
+        metric="x", mode="min")
+    return {"n_trials": len(analysis.trials),
+            "iters": [t.last_result["training_iteration"]
+                      for t in analysis.trials]}
+
+
+def _tune_train_fn(config):
+    from ray_lightning_amd import RayStrategy, Trainer
+    from ray_lightning_amd.tune import TuneReportCallback
+
+    trainer = Trainer(
+        max_epochs=config["max_epochs"], limit_train_batches=2,
+        limit_val_batches=1, num_sanity_val_steps=0,
+        default_root_dir=config["root"], enable_checkpointing=False,
+        callbacks=[TuneReportCallback(metrics={"x": "x"},
+                                      on="validation_end")],
+        strategy=RayStrategy(num_workers=1))
+    trainer.fit(BoringModel())
