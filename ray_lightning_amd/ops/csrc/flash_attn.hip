@@ -643,3 +643,248 @@ std::vector<torch::Tensor> flash_attn_bwd(
   TORCH_CHECK(e == hipSuccess, "flash_bwd: ", hipGetErrorString(e));
   return {dq, dk, dv};
 }
+
+// =====================================================================
+// forward v3 — swapped-operand S^T design (profiles/r01_flash_attn_notes
+// round-2 plan; guide T12 structure adapted to the verified 16x16x32
+// fragment maps):
+//   S^T = mfma(K, Q^T): C gives each lane 16 kv entries of ONE q column
+//   -> softmax over kv is lane-local + 2 cross-lane combines (vs 64
+//   shfl per tile in v2), and P^T feeds the PV B-operand after an
+//   in-register redistribution (no P LDS round-trip, no extra barrier).
+// Per workgroup: 64 q rows, 4 waves x 16 q columns; KV tiles of 64.
+// =====================================================================
+namespace {
+
+constexpr int WQ = 16;     // q columns per wave (v3)
+constexpr int BM3 = 64;    // q rows per workgroup (v3)
+
+__device__ __forceinline__ unsigned pack_bf16(float lo, float hi) {
+  union { unsigned u; __bf16 h[2]; } t;
+  t.h[0] = (__bf16)lo;
+  t.h[1] = (__bf16)hi;
+  return t.u;   // compiler emits v_cvt_pk_bf16_f32
+}
+
+// Redistribute the P^T C-layout (lane holds q = lane&15; kv =
+// 16*sub + 4*(lane>>4) + r) into the PV B-fragment layout (lane needs
+// 8 consecutive kv of its q column: kv = 32*blk + 8*(lane>>4) + i).
+// X_j / Y_j are the packed dwords of kv-subtiles 2*blk / 2*blk+1
+// (j = dword within the 4-run). Exchange is across the lane>>4 group
+// bits only (fixed per-register partner — the property that broke the
+// naive bpermute scheme, see r01 notes).
+//
+// Derivation (target group g needs source pack of group 2g mod 4,
+// register X for g<2 / Y for g>=2, dwords j then j of group 2g+1):
+//   D1 = {g0:X@g0, g1:X@g2, g2:Y@g0, g3:Y@g2}  -> B-frag dwords j
+//   D2 = {g0:X@g1, g1:X@g3, g2:Y@g1, g3:Y@g3}  -> B-frag dwords 2+j
+template <bool USE_PERMLANE>
+__device__ __forceinline__ void redist_pair(unsigned xj, unsigned yj,
+                                            int lane, unsigned& d1,
+                                            unsigned& d2) {
+  if (USE_PERMLANE) {
+    // (P,S) = pl32swap(X,Y): P = {g01: X own, g23: Y of g-2};
+    //                        S = {g01: X of g+2, g23: Y own}
+    // (D1,D2) = pl16swap(P,S): D1 = even 16-rows interleave, D2 = odd
+    auto ps = __builtin_amdgcn_permlane32_swap(xj, yj, false, false);
+    auto dd = __builtin_amdgcn_permlane16_swap(ps[0], ps[1], false,
+                                               false);
+    d1 = dd[0];
+    d2 = dd[1];
+  } else {
+    // ds_bpermute fallback (known-correct semantics): source lane =
+    // same q, group 2*(g&1) [+1 for d2]; register by target bit5.
+    const int src = (lane & 15) | ((lane & 16) << 1);
+    const unsigned x1 = __shfl(xj, src, 64);
+    const unsigned y1 = __shfl(yj, src, 64);
+    const unsigned x2 = __shfl(xj, src + 16, 64);
+    const unsigned y2 = __shfl(yj, src + 16, 64);
+    d1 = (lane & 32) ? y1 : x1;
+    d2 = (lane & 32) ? y2 : x2;
+  }
+}
+
+template <bool USE_PERMLANE>
+__global__ __launch_bounds__(256) void flash_fwd_v3_kernel(
+    const __hip_bfloat16* __restrict__ Q,
+    const __hip_bfloat16* __restrict__ K,
+    const __hip_bfloat16* __restrict__ V,
+    __hip_bfloat16* __restrict__ O, float* __restrict__ LSE,
+    int T, float scale) {
+  const int bh = blockIdx.y;
+  const int qm0 = blockIdx.x * BM3;
+  if (qm0 >= T) return;
+  const long base = (long)bh * T * HS;
+  const __hip_bfloat16* q = Q + base;
+  const __hip_bfloat16* k = K + base;
+  const __hip_bfloat16* v = V + base;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int q0 = qm0 + wave * WQ;      // wave's first (of 16) q column
+  const int myq = q0 + (lane & 15);    // this lane's q column
+  const int g = lane >> 4;             // kv group 0..3
+
+  __shared__ __bf16 lds_vt[2][HS * BN];   // V^T double buffer, 16 KB
+  __shared__ __bf16 lds_o[4][WQ * HS];    // epilogue transpose, 8 KB
+
+  // Q^T B-fragments (col = q = lane&15, k = hs): straight row-major
+  // 16-byte loads from Q (same bytes as an A-fragment of row q).
+  bf16x8 qf[2];
+  #pragma unroll
+  for (int kk = 0; kk < 2; ++kk)
+    qf[kk] = load_frag_rowmajor(q, q0, kk, lane, HS);
+
+  float m_i = kNegInf, l_i = 0.f;
+  f32x4 o_acc[4];   // O^T: row d = 16*n + 4*g + r, col q (lane&15)
+  #pragma unroll
+  for (int n = 0; n < 4; ++n)
+    o_acc[n] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const float l2e = 1.4426950408889634f * scale;
+  const int kv_end = qm0 + BM3;
+  int buf = 0;
+  for (int kn0 = 0; kn0 < kv_end && kn0 < T; kn0 += BN, buf ^= 1) {
+    stage_transposed(lds_vt[buf], v, kn0);
+    __syncthreads();
+    const bool active = kn0 <= q0 + WQ - 1;
+    if (active) {
+      // --- S^T = K Q^T: lane gets kv = kn0+16*sub+4*g+r of column myq
+      f32x4 st[4];
+      #pragma unroll
+      for (int sub = 0; sub < 4; ++sub) {
+        bf16x8 kf0 = load_frag_rowmajor(k, kn0 + 16 * sub, 0, lane, HS);
+        bf16x8 kf1 = load_frag_rowmajor(k, kn0 + 16 * sub, 1, lane, HS);
+        st[sub] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            kf0, qf[0], f32x4{0.f, 0.f, 0.f, 0.f}, 0, 0, 0);
+        st[sub] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            kf1, qf[1], st[sub], 0, 0, 0);
+      }
+      // --- causal mask (only tiles overlapping the diagonal) ---------
+      if (kn0 + BN - 1 > q0) {
+        #pragma unroll
+        for (int sub = 0; sub < 4; ++sub)
+          #pragma unroll
+          for (int r = 0; r < 4; ++r)
+            if (kn0 + 16 * sub + 4 * g + r > myq)
+              st[sub][r] = kNegInf;
+      }
+      // --- online softmax: lane-local over 16 kv + 2 combines --------
+      float mx = kNegInf;
+      #pragma unroll
+      for (int sub = 0; sub < 4; ++sub)
+        #pragma unroll
+        for (int r = 0; r < 4; ++r)
+          mx = fmaxf(mx, st[sub][r]);
+      mx = fmaxf(mx, __shfl_xor(mx, 16, 64));
+      mx = fmaxf(mx, __shfl_xor(mx, 32, 64));
+      const float m_new = fmaxf(m_i, mx);
+      const float corr = exp2f((m_i - m_new) * l2e);
+      float rowsum = 0.f;
+      #pragma unroll
+      for (int sub = 0; sub < 4; ++sub)
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const float p = (st[sub][r] <= kNegInf * 0.5f)
+              ? 0.f : exp2f((st[sub][r] - m_new) * l2e);
+          st[sub][r] = p;
+          rowsum += p;
+        }
+      rowsum += __shfl_xor(rowsum, 16, 64);
+      rowsum += __shfl_xor(rowsum, 32, 64);
+      l_i = l_i * corr + rowsum;
+      m_i = m_new;
+      #pragma unroll
+      for (int n = 0; n < 4; ++n)
+        #pragma unroll
+        for (int r = 0; r < 4; ++r)
+          o_acc[n][r] *= corr;
+      // --- P^T -> PV B-fragments, in registers -----------------------
+      bf16x8 pb[2];
+      #pragma unroll
+      for (int blk = 0; blk < 2; ++blk) {
+        const unsigned x0 = pack_bf16(st[2 * blk][0], st[2 * blk][1]);
+        const unsigned x1 = pack_bf16(st[2 * blk][2], st[2 * blk][3]);
+        const unsigned y0 = pack_bf16(st[2 * blk + 1][0],
+                                      st[2 * blk + 1][1]);
+        const unsigned y1 = pack_bf16(st[2 * blk + 1][2],
+                                      st[2 * blk + 1][3]);
+        unsigned d0, d2, d1, d3;
+        redist_pair<USE_PERMLANE>(x0, y0, lane, d0, d2);
+        redist_pair<USE_PERMLANE>(x1, y1, lane, d1, d3);
+        unsigned* pw = reinterpret_cast<unsigned*>(&pb[blk]);
+        pw[0] = d0; pw[1] = d1; pw[2] = d2; pw[3] = d3;
+      }
+      // --- O^T += V^T P^T -------------------------------------------
+      #pragma unroll
+      for (int n = 0; n < 4; ++n) {
+        #pragma unroll
+        for (int c = 0; c < 2; ++c) {
+          bf16x8 va;
+          *reinterpret_cast<int4*>(&va) =
+              *reinterpret_cast<const int4*>(
+                  lds_vt[buf] + (16 * n + (lane & 15)) * BN + 32 * c +
+                  g * 8);
+          o_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              va, pb[c], o_acc[n], 0, 0, 0);
+        }
+      }
+    }
+    // next tile writes the OTHER V^T buffer: no trailing barrier
+  }
+
+  // --- epilogue: transpose O^T -> O rows through LDS ------------------
+  const float inv_l = (l_i > 0.f) ? 1.f / l_i : 0.f;
+  __bf16* ow = lds_o[wave];
+  #pragma unroll
+  for (int n = 0; n < 4; ++n)
+    #pragma unroll
+    for (int r = 0; r < 4; ++r)
+      ow[(lane & 15) * HS + 16 * n + 4 * g + r] =
+          (__bf16)(o_acc[n][r] * inv_l);
+  wait_lds();
+  // 64 lanes x 2 b128: row = lane>>2 (16 rows), 16 bf16 per read
+  {
+    const int row = lane >> 2;           // wave-local q row 0..15
+    const int c0 = (lane & 3) * 16;
+    int4 t0 = *reinterpret_cast<const int4*>(ow + row * HS + c0);
+    int4 t1 = *reinterpret_cast<const int4*>(ow + row * HS + c0 + 8);
+    *reinterpret_cast<int4*>(O + base + (long)(q0 + row) * HS + c0) = t0;
+    *reinterpret_cast<int4*>(
+        O + base + (long)(q0 + row) * HS + c0 + 8) = t1;
+  }
+  if (g == 0)
+    LSE[(long)bh * T + myq] =
+        m_i * scale + logf(fmaxf(l_i, 1e-30f));
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> flash_attn_fwd_v3(torch::Tensor q,
+                                             torch::Tensor k,
+                                             torch::Tensor v,
+                                             double scale,
+                                             bool use_permlane) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16 &&
+              q.is_contiguous() && k.is_contiguous() &&
+              v.is_contiguous(), "flash_attn_fwd_v3: bf16 contiguous");
+  TORCH_CHECK(q.dim() == 4 && q.size(3) == HS,
+              "flash_attn_fwd_v3: [B,H,T,64] expected");
+  const int B = (int)q.size(0), H = (int)q.size(1), T = (int)q.size(2);
+  TORCH_CHECK(T % BM3 == 0, "flash_attn_fwd_v3: T % 64 == 0");
+  auto o = torch::empty_like(q);
+  auto lse = torch::empty({B, H, T}, q.options().dtype(at::kFloat));
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  dim3 grid(T / BM3, B * H);
+  auto* kern = use_permlane ? flash_fwd_v3_kernel<true>
+                            : flash_fwd_v3_kernel<false>;
+  hipLaunchKernelGGL(kern, grid, dim3(256), 0, stream,
+                     reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(k.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(v.data_ptr()),
+                     reinterpret_cast<__hip_bfloat16*>(o.data_ptr()),
+                     lse.data_ptr<float>(), T, (float)scale);
+  hipError_t e = hipGetLastError();
+  TORCH_CHECK(e == hipSuccess, "flash_fwd_v3: ", hipGetErrorString(e));
+  return {o, lse};
+}

@@ -645,10 +645,16 @@ std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q,
 std::vector<torch::Tensor> flash_attn_bwd(
     torch::Tensor dout, torch::Tensor q, torch::Tensor k,
     torch::Tensor v, torch::Tensor o, torch::Tensor lse, double scale);
+std::vector<torch::Tensor> flash_attn_fwd_v3(torch::Tensor q,
+                                             torch::Tensor k,
+                                             torch::Tensor v,
+                                             double scale,
+                                             bool use_permlane);
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
 torch::Tensor perm_probe(torch::Tensor M, torch::Tensor B,
                          long variant);
 std::vector<torch::Tensor> perm_dump(torch::Tensor M, long variant);
+torch::Tensor permlane_swap_probe();
 
 // fused residual-add + LayerNorm — defined in fused_ln.hip
 std::vector<torch::Tensor> fused_ln_fwd(
@@ -672,12 +678,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused residual-add + LayerNorm backward");
   m.def("flash_attn_fwd", &flash_attn_fwd,
         "causal flash attention forward (hs=64, bf16, MFMA)");
+  m.def("flash_attn_fwd_v3", &flash_attn_fwd_v3,
+        "swapped-operand MFMA flash attention fwd (v3)");
   m.def("flash_attn_bwd", &flash_attn_bwd,
         "causal flash attention backward");
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
   m.def("perm_probe", &perm_probe,
         "C-layout -> A-fragment bpermute redistribution probe");
   m.def("perm_dump", &perm_dump, "redistribution element dump");
+  m.def("permlane_swap_probe", &permlane_swap_probe,
+        "permlane{16,32}_swap lane-exchange semantics probe");
   m.def("scale_inplace", &scale_inplace, "flat *= s");
   m.def("scale_cast", &scale_cast, "dst_f32 = src_bf16 * s");
   m.def("fused_sgd", &fused_sgd, "fused multi-tensor SGD(momentum)");

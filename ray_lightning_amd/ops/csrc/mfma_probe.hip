@@ -246,3 +246,37 @@ std::vector<torch::Tensor> perm_dump(torch::Tensor M, long variant) {
                      (int)variant);
   return {out, ident};
 }
+
+namespace {
+// permlane{16,32}_swap semantics probe: a = 0x1000+lane, b = 0x2000+lane.
+// Row 0/1: permlane32_swap results (r0, r1); row 2/3: permlane16_swap.
+__global__ void permlane_swap_probe_kernel(unsigned* __restrict__ out) {
+  const unsigned lane = threadIdx.x;
+  if (lane >= 64) return;
+  const unsigned a = 0x1000u + lane;
+  const unsigned b = 0x2000u + lane;
+  {
+    auto r = __builtin_amdgcn_permlane32_swap(a, b, false, false);
+    out[lane] = r[0];
+    out[64 + lane] = r[1];
+  }
+  {
+    auto r = __builtin_amdgcn_permlane16_swap(a, b, false, false);
+    out[128 + lane] = r[0];
+    out[192 + lane] = r[1];
+  }
+}
+}  // namespace
+
+torch::Tensor permlane_swap_probe() {
+  auto out = torch::zeros({4, 64}, torch::dtype(at::kInt)
+                                       .device(at::kCUDA));
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(permlane_swap_probe_kernel, dim3(1), dim3(64), 0,
+                     stream,
+                     reinterpret_cast<unsigned*>(out.data_ptr()));
+  hipError_t e = hipGetLastError();
+  TORCH_CHECK(e == hipSuccess, "permlane_swap_probe: ",
+              hipGetErrorString(e));
+  return out;
+}

@@ -32,6 +32,9 @@ def t(fn, n=20):
 
 o, lse = ext.flash_attn_fwd(q, k, v, scale)
 t_fwd = t(lambda: ext.flash_attn_fwd(q, k, v, scale))
+t_fwd3s = t(lambda: ext.flash_attn_fwd_v3(q, k, v, scale, False))
+t_fwd3p = t(lambda: ext.flash_attn_fwd_v3(q, k, v, scale, True))
+print(f"fwd v3(shfl) {t_fwd3s:.3f} ms   fwd v3(permlane) {t_fwd3p:.3f} ms")
 t_bwd = t(lambda: ext.flash_attn_bwd(dy, q, k, v, o, lse, scale))
 print(f"fwd {t_fwd:.3f} ms   bwd(all3) {t_bwd:.3f} ms")
 

@@ -151,3 +151,44 @@ def test_perm_redistribution_probe():
         errs.append(float((C2 - ref).abs().max()))
     print(f"\n[perm_probe] variant errs: {errs}")
     assert min(errs) < 3e-2, f"neither perm variant matches: {errs}"
+
+
+@pytest.mark.parametrize("use_permlane", [False, True])
+@pytest.mark.parametrize("B,H,T", [(2, 3, 256), (1, 2, 1024),
+                                   (1, 1, 64)])
+def test_flash_fwd_v3_numerics(B, H, T, use_permlane):
+    """v3 (swapped-operand S^T, in-register P redistribution) against
+    the fp32 reference; both redistribution paths."""
+    from ray_lightning_amd import ops
+    ext = ops._load_ext()
+    torch.manual_seed(0)
+    hs = 64
+    q = (torch.randn(B, H, T, hs, device="cuda") * 0.5).bfloat16()
+    k = (torch.randn(B, H, T, hs, device="cuda") * 0.5).bfloat16()
+    v = (torch.randn(B, H, T, hs, device="cuda") * 0.5).bfloat16()
+    scale = 1.0 / math.sqrt(hs)
+    o, lse = ext.flash_attn_fwd_v3(q, k, v, scale, use_permlane)
+    ref = _ref_attention(q, k, v, scale)
+    assert torch.allclose(o.float(), ref, atol=3e-2, rtol=3e-2), \
+        f"max err {(o.float() - ref).abs().max()}"
+    # LSE against the fp32 reference logsumexp
+    s = (q.float() @ k.float().transpose(-1, -2)) * scale
+    mask = torch.tril(
+        torch.ones(T, T, device=q.device, dtype=torch.bool))
+    s = s.masked_fill(~mask, float("-inf"))
+    lse_ref = torch.logsumexp(s, dim=-1)
+    assert torch.allclose(lse, lse_ref, atol=2e-3, rtol=1e-3), \
+        f"LSE max err {(lse - lse_ref).abs().max()}"
+
+
+def test_flash_fwd_v3_matches_v2_lse():
+    from ray_lightning_amd import ops
+    ext = ops._load_ext()
+    torch.manual_seed(1)
+    q = (torch.randn(2, 2, 512, 64, device="cuda")).bfloat16()
+    k = torch.randn_like(q).bfloat16()
+    v = torch.randn_like(q).bfloat16()
+    o2, lse2 = ext.flash_attn_fwd(q, k, v, 0.125)
+    o3, lse3 = ext.flash_attn_fwd_v3(q, k, v, 0.125, False)
+    assert torch.allclose(lse2, lse3, atol=2e-3, rtol=1e-3)
+    assert torch.allclose(o2.float(), o3.float(), atol=2e-2, rtol=2e-2)
