@@ -73,6 +73,54 @@ __device__ __forceinline__ void stage_transposed(
   }
 }
 
+
+// --- XOR-swizzled transpose image (v3) --------------------------------
+// Image: img[d][kv] bf16, row = 128 B. Bank index depends on byte%256
+// (two rows), so consecutive d alternate half-banks; the granule sel
+// (((d>>1)^(d>>3))&7) spreads the 8 same-parity rows of a 16-row
+// fragment read over all 8 granules -> conflict-free ds_read_b128,
+// and distinct granules for the d / d+8 write groups.
+__device__ __forceinline__ int swz_off(int d, int byte_in_row) {
+  return d * (2 * BN) +
+         (byte_in_row ^ ((((d >> 1) ^ (d >> 3)) & 7) << 4));
+}
+
+// Transpose-stage a [BN, HS] row-major global tile into img[HS][BN]
+// with dword-paired writes (two rows per ds_write_b32: 8 b32 writes
+// per thread at <=2-way instead of 16 scalar b16 at up to 8-way).
+__device__ __forceinline__ void stage_transposed_swz(
+    __bf16* dst, const __hip_bfloat16* src, int row0) {
+  const int p = threadIdx.x & 31;        // row pair: rows 2p, 2p+1
+  const int c0 = (threadIdx.x >> 5) * 8; // 8-column run
+  bf16x8 t0, t1;
+  *reinterpret_cast<int4*>(&t0) = *reinterpret_cast<const int4*>(
+      src + (long)(row0 + 2 * p) * HS + c0);
+  *reinterpret_cast<int4*>(&t1) = *reinterpret_cast<const int4*>(
+      src + (long)(row0 + 2 * p + 1) * HS + c0);
+  char* base = reinterpret_cast<char*>(dst);
+  #pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    union { unsigned u; __bf16 h[2]; } w;
+    w.h[0] = t0[i];
+    w.h[1] = t1[i];
+    *reinterpret_cast<unsigned*>(
+        base + swz_off(c0 + i, 4 * p)) = w.u;
+  }
+}
+
+// b128 fragment read from the swizzled image: row = row16 + (lane&15),
+// bytes [16*(4c+... ) fixed 16-byte granule at (32c+8g)*2.
+__device__ __forceinline__ bf16x8 read_frag_swz(const __bf16* img,
+                                                int row16, int c,
+                                                int lane) {
+  const int d = row16 + (lane & 15);
+  const int g = lane >> 4;
+  bf16x8 v;
+  *reinterpret_cast<int4*>(&v) = *reinterpret_cast<const int4*>(
+      reinterpret_cast<const char*>(img) + swz_off(d, 64 * c + 16 * g));
+  return v;
+}
+
 // ---------------------------------------------------------------------
 // forward
 // ---------------------------------------------------------------------
@@ -354,8 +402,8 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
   const float l2e = 1.4426950408889634f;
 
   for (int qm0 = kb0; qm0 < T; qm0 += BN) {  // q tiles of 64 rows
-    stage_transposed(lds_dot, dO, qm0);
-    stage_transposed(lds_qt, q, qm0);
+    stage_transposed_swz(lds_dot, dO, qm0);
+    stage_transposed_swz(lds_qt, q, qm0);
     if (threadIdx.x < BN) {
       lds_lse[threadIdx.x] = lse[qm0 + threadIdx.x];
       lds_d[threadIdx.x] = Drow[qm0 + threadIdx.x];
@@ -745,7 +793,7 @@ __global__ __launch_bounds__(256) void flash_fwd_v3_kernel(
   const int kv_end = qm0 + BM3;
   int buf = 0;
   for (int kn0 = 0; kn0 < kv_end && kn0 < T; kn0 += BN, buf ^= 1) {
-    stage_transposed(lds_vt[buf], v, kn0);
+    stage_transposed_swz(lds_vt[buf], v, kn0);
     __syncthreads();
     const bool active = kn0 <= q0 + WQ - 1;
     if (active) {
@@ -820,11 +868,7 @@ __global__ __launch_bounds__(256) void flash_fwd_v3_kernel(
       for (int n = 0; n < 4; ++n) {
         #pragma unroll
         for (int c = 0; c < 2; ++c) {
-          bf16x8 va;
-          *reinterpret_cast<int4*>(&va) =
-              *reinterpret_cast<const int4*>(
-                  lds_vt[buf] + (16 * n + (lane & 15)) * BN + 32 * c +
-                  g * 8);
+          bf16x8 va = read_frag_swz(lds_vt[buf], 16 * n, c, lane);
           o_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               va, pb[c], o_acc[n], 0, 0, 0);
         }
@@ -887,4 +931,342 @@ std::vector<torch::Tensor> flash_attn_fwd_v3(torch::Tensor q,
   hipError_t e = hipGetLastError();
   TORCH_CHECK(e == hipSuccess, "flash_fwd_v3: ", hipGetErrorString(e));
   return {o, lse};
+}
+
+// =====================================================================
+// backward v3 — same swapped-C-layout + in-register redistribution
+// structure as fwd v3 (no P^T/dS^T LDS round-trips, no per-element
+// softmax shuffles; lse/D land lane-resident where the layout allows).
+// =====================================================================
+namespace {
+
+// dK/dV v3: workgroup = 64 KV rows, 4 waves x 16 kv columns; iterate
+// q tiles of 64. C layouts put kv in lane&15 throughout:
+//   S[q][kv]   = mfma(Q_frag,  K^T B-frag)   (both direct row-major)
+//   dP[q][kv]  = mfma(dO_frag, V^T B-frag)
+//   dV^T[d][kv]= mfma(dO^T A-frag(lds), redist(P) B-frag)
+//   dK^T[d][kv]= mfma(Q^T  A-frag(lds), redist(dS) B-frag)
+template <bool USE_PERMLANE>
+__global__ __launch_bounds__(256) void flash_bwd_dkv_v3_kernel(
+    const __hip_bfloat16* __restrict__ Qg,
+    const __hip_bfloat16* __restrict__ Kg,
+    const __hip_bfloat16* __restrict__ Vg,
+    const __hip_bfloat16* __restrict__ dOg,
+    const float* __restrict__ LSE, const float* __restrict__ Dg,
+    __hip_bfloat16* __restrict__ dK, __hip_bfloat16* __restrict__ dV,
+    int T, float scale) {
+  const int bh = blockIdx.y;
+  const int kb0 = blockIdx.x * BM3;      // 64 KV rows per workgroup
+  if (kb0 >= T) return;
+  const long base = (long)bh * T * HS;
+  const __hip_bfloat16* q = Qg + base;
+  const __hip_bfloat16* k = Kg + base;
+  const __hip_bfloat16* v = Vg + base;
+  const __hip_bfloat16* dO = dOg + base;
+  const float* lse = LSE + (long)bh * T;
+  const float* Drow = Dg + (long)bh * T;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int kv0 = kb0 + wave * WQ;       // wave's first kv column
+  const int mykv = kv0 + (lane & 15);
+  const int g = lane >> 4;
+
+  __shared__ __bf16 lds_dot[HS * BN];    // dO^T [hs][q]  8 KB
+  __shared__ __bf16 lds_qt[HS * BN];     // Q^T  [hs][q]  8 KB
+  __shared__ float lds_lse[BN];
+  __shared__ float lds_d[BN];
+  __shared__ __bf16 lds_ep[4][WQ * HS];  // epilogue     8 KB
+
+  // K^T / V^T B-fragments (col = kv = lane&15): kernel-resident
+  bf16x8 kf[2], vf[2];
+  #pragma unroll
+  for (int kk = 0; kk < 2; ++kk) {
+    kf[kk] = load_frag_rowmajor(k, kv0, kk, lane, HS);
+    vf[kk] = load_frag_rowmajor(v, kv0, kk, lane, HS);
+  }
+
+  f32x4 dv_acc[4], dk_acc[4];   // ^T: row d = 16n+4g+r, col kv
+  #pragma unroll
+  for (int n = 0; n < 4; ++n) {
+    dv_acc[n] = f32x4{0.f, 0.f, 0.f, 0.f};
+    dk_acc[n] = f32x4{0.f, 0.f, 0.f, 0.f};
+  }
+
+  const float l2e = 1.4426950408889634f;
+
+  for (int qm0 = kb0; qm0 < T; qm0 += BN) {
+    stage_transposed_swz(lds_dot, dO, qm0);
+    stage_transposed_swz(lds_qt, q, qm0);
+    if (threadIdx.x < BN) {
+      lds_lse[threadIdx.x] = lse[qm0 + threadIdx.x];
+      lds_d[threadIdx.x] = Drow[qm0 + threadIdx.x];
+    }
+    __syncthreads();
+    const bool active = kv0 <= qm0 + BN - 1;
+    if (active) {
+      float pv[4][4], dsv[4][4];
+      #pragma unroll
+      for (int sub = 0; sub < 4; ++sub) {
+        bf16x8 aq0 = load_frag_rowmajor(q, qm0 + 16 * sub, 0, lane, HS);
+        bf16x8 aq1 = load_frag_rowmajor(q, qm0 + 16 * sub, 1, lane, HS);
+        bf16x8 ad0 = load_frag_rowmajor(dO, qm0 + 16 * sub, 0, lane,
+                                        HS);
+        bf16x8 ad1 = load_frag_rowmajor(dO, qm0 + 16 * sub, 1, lane,
+                                        HS);
+        f32x4 sacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            aq0, kf[0], f32x4{0.f, 0.f, 0.f, 0.f}, 0, 0, 0);
+        sacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            aq1, kf[1], sacc, 0, 0, 0);
+        f32x4 dpacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            ad0, vf[0], f32x4{0.f, 0.f, 0.f, 0.f}, 0, 0, 0);
+        dpacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            ad1, vf[1], dpacc, 0, 0, 0);
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int qi = 16 * sub + 4 * g + r;     // tile-local q
+          const bool valid = (qm0 + qi) >= mykv;
+          const float p = valid
+              ? exp2f((scale * sacc[r] - lds_lse[qi]) * l2e) : 0.f;
+          pv[sub][r] = p;
+          dsv[sub][r] = valid
+              ? scale * p * (dpacc[r] - lds_d[qi]) : 0.f;
+        }
+      }
+      // redistribute over the q axis -> B-fragments (k = q)
+      #pragma unroll
+      for (int blk = 0; blk < 2; ++blk) {
+        bf16x8 pb, db;
+        {
+          const unsigned x0 = pack_bf16(pv[2 * blk][0], pv[2 * blk][1]);
+          const unsigned x1 = pack_bf16(pv[2 * blk][2], pv[2 * blk][3]);
+          const unsigned y0 = pack_bf16(pv[2 * blk + 1][0],
+                                        pv[2 * blk + 1][1]);
+          const unsigned y1 = pack_bf16(pv[2 * blk + 1][2],
+                                        pv[2 * blk + 1][3]);
+          unsigned d0, d1, d2, d3;
+          redist_pair<USE_PERMLANE>(x0, y0, lane, d0, d2);
+          redist_pair<USE_PERMLANE>(x1, y1, lane, d1, d3);
+          unsigned* pw = reinterpret_cast<unsigned*>(&pb);
+          pw[0] = d0; pw[1] = d1; pw[2] = d2; pw[3] = d3;
+        }
+        {
+          const unsigned x0 = pack_bf16(dsv[2 * blk][0],
+                                        dsv[2 * blk][1]);
+          const unsigned x1 = pack_bf16(dsv[2 * blk][2],
+                                        dsv[2 * blk][3]);
+          const unsigned y0 = pack_bf16(dsv[2 * blk + 1][0],
+                                        dsv[2 * blk + 1][1]);
+          const unsigned y1 = pack_bf16(dsv[2 * blk + 1][2],
+                                        dsv[2 * blk + 1][3]);
+          unsigned d0, d1, d2, d3;
+          redist_pair<USE_PERMLANE>(x0, y0, lane, d0, d2);
+          redist_pair<USE_PERMLANE>(x1, y1, lane, d1, d3);
+          unsigned* pw = reinterpret_cast<unsigned*>(&db);
+          pw[0] = d0; pw[1] = d1; pw[2] = d2; pw[3] = d3;
+        }
+        #pragma unroll
+        for (int n = 0; n < 4; ++n) {
+          bf16x8 adot = read_frag_swz(lds_dot, 16 * n, blk, lane);
+          bf16x8 aqt = read_frag_swz(lds_qt, 16 * n, blk, lane);
+          dv_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              adot, pb, dv_acc[n], 0, 0, 0);
+          dk_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              aqt, db, dk_acc[n], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  // epilogue: transpose dV^T then dK^T through per-wave LDS
+  __bf16* ew = lds_ep[wave];
+  const int erow = lane >> 2;            // wave-local kv row
+  const int ec0 = (lane & 3) * 16;
+  #pragma unroll
+  for (int which = 0; which < 2; ++which) {
+    f32x4* acc = which ? dk_acc : dv_acc;
+    __hip_bfloat16* out = which ? dK : dV;
+    #pragma unroll
+    for (int n = 0; n < 4; ++n)
+      #pragma unroll
+      for (int r = 0; r < 4; ++r)
+        ew[(lane & 15) * HS + 16 * n + 4 * g + r] =
+            (__bf16)acc[n][r];
+    wait_lds();
+    int4 t0 = *reinterpret_cast<const int4*>(ew + erow * HS + ec0);
+    int4 t1 = *reinterpret_cast<const int4*>(ew + erow * HS + ec0 + 8);
+    *reinterpret_cast<int4*>(
+        out + base + (long)(kv0 + erow) * HS + ec0) = t0;
+    *reinterpret_cast<int4*>(
+        out + base + (long)(kv0 + erow) * HS + ec0 + 8) = t1;
+    wait_lds();
+  }
+}
+
+// dQ v3: workgroup = 64 q rows, 4 waves x 16 q columns; iterate kv
+// tiles of 64. lse/D are lane-resident (q fixed per lane).
+//   S^T[kv][q]  = mfma(K_frag,  Q^T B-frag)
+//   dP^T[kv][q] = mfma(V_frag,  dO^T B-frag)
+//   dQ^T[d][q]  = mfma(K^T A-frag(lds), redist(dS^T) B-frag)
+template <bool USE_PERMLANE>
+__global__ __launch_bounds__(256) void flash_bwd_dq_v3_kernel(
+    const __hip_bfloat16* __restrict__ Qg,
+    const __hip_bfloat16* __restrict__ Kg,
+    const __hip_bfloat16* __restrict__ Vg,
+    const __hip_bfloat16* __restrict__ dOg,
+    const float* __restrict__ LSE, const float* __restrict__ Dg,
+    __hip_bfloat16* __restrict__ dQ, int T, float scale) {
+  const int bh = blockIdx.y;
+  const int qm0 = blockIdx.x * BM3;
+  if (qm0 >= T) return;
+  const long base = (long)bh * T * HS;
+  const __hip_bfloat16* q = Qg + base;
+  const __hip_bfloat16* k = Kg + base;
+  const __hip_bfloat16* v = Vg + base;
+  const __hip_bfloat16* dO = dOg + base;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int q0 = qm0 + wave * WQ;
+  const int myq = q0 + (lane & 15);
+  const int g = lane >> 4;
+
+  __shared__ __bf16 lds_kt[HS * BN];     // K^T [hs][kv]  8 KB
+  __shared__ __bf16 lds_ep[4][WQ * HS];  // epilogue      8 KB
+
+  bf16x8 qf[2], dof[2];
+  #pragma unroll
+  for (int kk = 0; kk < 2; ++kk) {
+    qf[kk] = load_frag_rowmajor(q, q0, kk, lane, HS);
+    dof[kk] = load_frag_rowmajor(dO, q0, kk, lane, HS);
+  }
+  const float lse_q = LSE[(long)bh * T + myq];
+  const float d_q = Dg[(long)bh * T + myq];
+
+  f32x4 dq_acc[4];    // dQ^T: row d, col q
+  #pragma unroll
+  for (int n = 0; n < 4; ++n)
+    dq_acc[n] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const float l2e = 1.4426950408889634f;
+  const int kv_end = qm0 + BM3;
+  for (int kn0 = 0; kn0 < kv_end && kn0 < T; kn0 += BN) {
+    stage_transposed_swz(lds_kt, k, kn0);
+    __syncthreads();
+    const bool active = kn0 <= q0 + WQ - 1;
+    if (active) {
+      float dsv[4][4];
+      #pragma unroll
+      for (int sub = 0; sub < 4; ++sub) {
+        bf16x8 ak0 = load_frag_rowmajor(k, kn0 + 16 * sub, 0, lane, HS);
+        bf16x8 ak1 = load_frag_rowmajor(k, kn0 + 16 * sub, 1, lane, HS);
+        bf16x8 av0 = load_frag_rowmajor(v, kn0 + 16 * sub, 0, lane, HS);
+        bf16x8 av1 = load_frag_rowmajor(v, kn0 + 16 * sub, 1, lane, HS);
+        f32x4 st = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            ak0, qf[0], f32x4{0.f, 0.f, 0.f, 0.f}, 0, 0, 0);
+        st = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            ak1, qf[1], st, 0, 0, 0);
+        f32x4 dpt = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            av0, dof[0], f32x4{0.f, 0.f, 0.f, 0.f}, 0, 0, 0);
+        dpt = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            av1, dof[1], dpt, 0, 0, 0);
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int kvr = kn0 + 16 * sub + 4 * g + r;
+          const bool valid = kvr <= myq;
+          const float p = valid
+              ? exp2f((scale * st[r] - lse_q) * l2e) : 0.f;
+          dsv[sub][r] = valid
+              ? scale * p * (dpt[r] - d_q) : 0.f;
+        }
+      }
+      #pragma unroll
+      for (int blk = 0; blk < 2; ++blk) {
+        bf16x8 db;
+        const unsigned x0 = pack_bf16(dsv[2 * blk][0], dsv[2 * blk][1]);
+        const unsigned x1 = pack_bf16(dsv[2 * blk][2], dsv[2 * blk][3]);
+        const unsigned y0 = pack_bf16(dsv[2 * blk + 1][0],
+                                      dsv[2 * blk + 1][1]);
+        const unsigned y1 = pack_bf16(dsv[2 * blk + 1][2],
+                                      dsv[2 * blk + 1][3]);
+        unsigned d0, d1, d2, d3;
+        redist_pair<USE_PERMLANE>(x0, y0, lane, d0, d2);
+        redist_pair<USE_PERMLANE>(x1, y1, lane, d1, d3);
+        unsigned* pw = reinterpret_cast<unsigned*>(&db);
+        pw[0] = d0; pw[1] = d1; pw[2] = d2; pw[3] = d3;
+        #pragma unroll
+        for (int n = 0; n < 4; ++n) {
+          bf16x8 akt = read_frag_swz(lds_kt, 16 * n, blk, lane);
+          dq_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              akt, db, dq_acc[n], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  // epilogue transpose
+  __bf16* ew = lds_ep[wave];
+  #pragma unroll
+  for (int n = 0; n < 4; ++n)
+    #pragma unroll
+    for (int r = 0; r < 4; ++r)
+      ew[(lane & 15) * HS + 16 * n + 4 * g + r] = (__bf16)dq_acc[n][r];
+  wait_lds();
+  const int erow = lane >> 2;
+  const int ec0 = (lane & 3) * 16;
+  int4 t0 = *reinterpret_cast<const int4*>(ew + erow * HS + ec0);
+  int4 t1 = *reinterpret_cast<const int4*>(ew + erow * HS + ec0 + 8);
+  *reinterpret_cast<int4*>(
+      dQ + base + (long)(q0 + erow) * HS + ec0) = t0;
+  *reinterpret_cast<int4*>(
+      dQ + base + (long)(q0 + erow) * HS + ec0 + 8) = t1;
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> flash_attn_bwd_v3(
+    torch::Tensor dout, torch::Tensor q, torch::Tensor k,
+    torch::Tensor v, torch::Tensor o, torch::Tensor lse, double scale,
+    bool use_permlane) {
+  const int B = (int)q.size(0), H = (int)q.size(1), T = (int)q.size(2);
+  TORCH_CHECK(T % BM3 == 0, "flash_attn_bwd_v3: T % 64 == 0");
+  dout = dout.contiguous();
+  auto dq = torch::empty_like(q);
+  auto dk = torch::empty_like(k);
+  auto dv = torch::empty_like(v);
+  auto D = torch::empty({B, H, T}, q.options().dtype(at::kFloat));
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  const long total_rows = (long)B * H * T;
+  hipLaunchKernelGGL(flash_bwd_pre_kernel,
+                     dim3((total_rows + 63) / 64), dim3(256), 0, stream,
+                     reinterpret_cast<const __hip_bfloat16*>(dout.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(o.data_ptr()),
+                     D.data_ptr<float>(), total_rows);
+  dim3 grid(T / BM3, B * H);
+  auto* kdkv = use_permlane ? flash_bwd_dkv_v3_kernel<true>
+                            : flash_bwd_dkv_v3_kernel<false>;
+  auto* kdq = use_permlane ? flash_bwd_dq_v3_kernel<true>
+                           : flash_bwd_dq_v3_kernel<false>;
+  hipLaunchKernelGGL(kdkv, grid, dim3(256), 0, stream,
+                     reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(k.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(v.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(dout.data_ptr()),
+                     lse.data_ptr<float>(), D.data_ptr<float>(),
+                     reinterpret_cast<__hip_bfloat16*>(dk.data_ptr()),
+                     reinterpret_cast<__hip_bfloat16*>(dv.data_ptr()),
+                     T, (float)scale);
+  hipLaunchKernelGGL(kdq, grid, dim3(256), 0, stream,
+                     reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(k.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(v.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(dout.data_ptr()),
+                     lse.data_ptr<float>(), D.data_ptr<float>(),
+                     reinterpret_cast<__hip_bfloat16*>(dq.data_ptr()),
+                     T, (float)scale);
+  hipError_t e = hipGetLastError();
+  TORCH_CHECK(e == hipSuccess, "flash_bwd_v3: ", hipGetErrorString(e));
+  return {dq, dk, dv};
 }

@@ -650,6 +650,10 @@ std::vector<torch::Tensor> flash_attn_fwd_v3(torch::Tensor q,
                                              torch::Tensor v,
                                              double scale,
                                              bool use_permlane);
+std::vector<torch::Tensor> flash_attn_bwd_v3(
+    torch::Tensor dout, torch::Tensor q, torch::Tensor k,
+    torch::Tensor v, torch::Tensor o, torch::Tensor lse, double scale,
+    bool use_permlane);
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
 torch::Tensor perm_probe(torch::Tensor M, torch::Tensor B,
                          long variant);
@@ -682,6 +686,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "swapped-operand MFMA flash attention fwd (v3)");
   m.def("flash_attn_bwd", &flash_attn_bwd,
         "causal flash attention backward");
+  m.def("flash_attn_bwd_v3", &flash_attn_bwd_v3,
+        "swapped-operand MFMA flash attention bwd (v3)");
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
   m.def("perm_probe", &perm_probe,
         "C-layout -> A-fragment bpermute redistribution probe");

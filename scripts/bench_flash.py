@@ -36,7 +36,10 @@ t_fwd3s = t(lambda: ext.flash_attn_fwd_v3(q, k, v, scale, False))
 t_fwd3p = t(lambda: ext.flash_attn_fwd_v3(q, k, v, scale, True))
 print(f"fwd v3(shfl) {t_fwd3s:.3f} ms   fwd v3(permlane) {t_fwd3p:.3f} ms")
 t_bwd = t(lambda: ext.flash_attn_bwd(dy, q, k, v, o, lse, scale))
+t_bwd3s = t(lambda: ext.flash_attn_bwd_v3(dy, q, k, v, o, lse, scale, False))
+t_bwd3p = t(lambda: ext.flash_attn_bwd_v3(dy, q, k, v, o, lse, scale, True))
 print(f"fwd {t_fwd:.3f} ms   bwd(all3) {t_bwd:.3f} ms")
+print(f"bwd v3(shfl) {t_bwd3s:.3f} ms   bwd v3(permlane) {t_bwd3p:.3f} ms")
 
 # SDPA split
 qs = q.clone().requires_grad_(True)
