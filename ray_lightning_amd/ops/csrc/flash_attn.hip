@@ -90,8 +90,9 @@ __device__ __forceinline__ int swz_off(int d, int byte_in_row) {
 // per thread at <=2-way instead of 16 scalar b16 at up to 8-way).
 __device__ __forceinline__ void stage_transposed_swz(
     __bf16* dst, const __hip_bfloat16* src, int row0) {
-  const int p = threadIdx.x & 31;        // row pair: rows 2p, 2p+1
-  const int c0 = (threadIdx.x >> 5) * 8; // 8-column run
+  const int p = threadIdx.x >> 3;        // row pair: rows 2p, 2p+1
+  const int c0 = (threadIdx.x & 7) * 8;  // 8-column run (8 consecutive
+  // lanes cover one 128 B source row: coalesced global loads)
   bf16x8 t0, t1;
   *reinterpret_cast<int4*>(&t0) = *reinterpret_cast<const int4*>(
       src + (long)(row0 + 2 * p) * HS + c0);
@@ -402,8 +403,8 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
   const float l2e = 1.4426950408889634f;
 
   for (int qm0 = kb0; qm0 < T; qm0 += BN) {  // q tiles of 64 rows
-    stage_transposed_swz(lds_dot, dO, qm0);
-    stage_transposed_swz(lds_qt, q, qm0);
+    stage_transposed(lds_dot, dO, qm0);
+    stage_transposed(lds_qt, q, qm0);
     if (threadIdx.x < BN) {
       lds_lse[threadIdx.x] = lse[qm0 + threadIdx.x];
       lds_d[threadIdx.x] = Drow[qm0 + threadIdx.x];
