@@ -109,6 +109,70 @@ __device__ __forceinline__ void stage_transposed_swz(
   }
 }
 
+// Split staging (T14 issue-early/write-late): load the next tile's
+// piece into registers at the top of the current tile's compute, write
+// it to the other LDS buffer after the compute — HBM latency hides
+// under the MFMA/softmax phase instead of sitting in front of the
+// barrier.
+struct StageRegsT {   // V^T transpose staging piece (2 source rows)
+  bf16x8 r0, r1;
+};
+
+__device__ __forceinline__ StageRegsT stage_t_load(
+    const __hip_bfloat16* src, int row0) {
+  const int p = threadIdx.x >> 3;
+  const int c0 = (threadIdx.x & 7) * 8;
+  StageRegsT t;
+  *reinterpret_cast<int4*>(&t.r0) = *reinterpret_cast<const int4*>(
+      src + (long)(row0 + 2 * p) * HS + c0);
+  *reinterpret_cast<int4*>(&t.r1) = *reinterpret_cast<const int4*>(
+      src + (long)(row0 + 2 * p + 1) * HS + c0);
+  return t;
+}
+
+__device__ __forceinline__ void stage_t_write(__bf16* dst,
+                                              const StageRegsT& t) {
+  const int p = threadIdx.x >> 3;
+  const int c0 = (threadIdx.x & 7) * 8;
+  char* base = reinterpret_cast<char*>(dst);
+  #pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    union { unsigned u; __bf16 h[2]; } w;
+    w.h[0] = t.r0[i];
+    w.h[1] = t.r1[i];
+    *reinterpret_cast<unsigned*>(base + swz_off(c0 + i, 4 * p)) = w.u;
+  }
+}
+
+struct StageRegsR {   // row-major staging piece (rows r, r+32)
+  bf16x8 r0, r1;
+};
+
+__device__ __forceinline__ StageRegsR stage_r_load(
+    const __hip_bfloat16* src, int row0) {
+  const int r = threadIdx.x >> 3;
+  const int c0 = (threadIdx.x & 7) * 8;
+  StageRegsR t;
+  *reinterpret_cast<int4*>(&t.r0) = *reinterpret_cast<const int4*>(
+      src + (long)(row0 + r) * HS + c0);
+  *reinterpret_cast<int4*>(&t.r1) = *reinterpret_cast<const int4*>(
+      src + (long)(row0 + r + 32) * HS + c0);
+  return t;
+}
+
+// Row-major swizzled image (for A-fragment reads via read_frag_swz):
+// img[row][hs], b128 writes land in one granule each.
+__device__ __forceinline__ void stage_r_write(__bf16* dst,
+                                              const StageRegsR& t) {
+  const int r = threadIdx.x >> 3;
+  const int c0 = (threadIdx.x & 7) * 8;
+  char* base = reinterpret_cast<char*>(dst);
+  *reinterpret_cast<int4*>(base + swz_off(r, c0 * 2)) =
+      *reinterpret_cast<const int4*>(&t.r0);
+  *reinterpret_cast<int4*>(base + swz_off(r + 32, c0 * 2)) =
+      *reinterpret_cast<const int4*>(&t.r1);
+}
+
 // b128 fragment read from the swizzled image: row = row16 + (lane&15),
 // bytes [16*(4c+... ) fixed 16-byte granule at (32c+8g)*2.
 __device__ __forceinline__ bf16x8 read_frag_swz(const __bf16* img,
@@ -754,7 +818,9 @@ __device__ __forceinline__ void redist_pair(unsigned xj, unsigned yj,
 }
 
 template <bool USE_PERMLANE>
-__global__ __launch_bounds__(256) void flash_fwd_v3_kernel(
+// min 4 waves/EU: keeps the allocator at <=128 regs so occupancy stays
+// at the LDS-allowed 4 waves/SIMD (131 regs would drop it to 3)
+__global__ __launch_bounds__(256, 4) void flash_fwd_v3_kernel(
     const __hip_bfloat16* __restrict__ Q,
     const __hip_bfloat16* __restrict__ K,
     const __hip_bfloat16* __restrict__ V,
@@ -775,6 +841,7 @@ __global__ __launch_bounds__(256) void flash_fwd_v3_kernel(
   const int g = lane >> 4;             // kv group 0..3
 
   __shared__ __bf16 lds_vt[2][HS * BN];   // V^T double buffer, 16 KB
+  __shared__ __bf16 lds_k[2][BN * HS];    // K row-major db buffer, 16 KB
   __shared__ __bf16 lds_o[4][WQ * HS];    // epilogue transpose, 8 KB
 
   // Q^T B-fragments (col = q = lane&15, k = hs): straight row-major
@@ -791,19 +858,29 @@ __global__ __launch_bounds__(256) void flash_fwd_v3_kernel(
     o_acc[n] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const float l2e = 1.4426950408889634f * scale;
-  const int kv_end = qm0 + BM3;
+  const int kv_end = (qm0 + BM3 < T) ? qm0 + BM3 : T;
+  // prologue: tile 0 staged through registers, then the loop keeps one
+  // tile in flight (issue loads -> compute current -> write other buf)
+  {
+    StageRegsT vt = stage_t_load(v, 0);
+    StageRegsR kt = stage_r_load(k, 0);
+    stage_t_write(lds_vt[0], vt);
+    stage_r_write(lds_k[0], kt);
+  }
+  __syncthreads();
   int buf = 0;
-  for (int kn0 = 0; kn0 < kv_end && kn0 < T; kn0 += BN, buf ^= 1) {
-    stage_transposed_swz(lds_vt[buf], v, kn0);
-    __syncthreads();
+  for (int kn0 = 0; kn0 < kv_end; kn0 += BN, buf ^= 1) {
+    const int next = (kn0 + BN < kv_end) ? kn0 + BN : kn0;
+    StageRegsT vt = stage_t_load(v, next);
+    StageRegsR kt = stage_r_load(k, next);
     const bool active = kn0 <= q0 + WQ - 1;
     if (active) {
       // --- S^T = K Q^T: lane gets kv = kn0+16*sub+4*g+r of column myq
       f32x4 st[4];
       #pragma unroll
       for (int sub = 0; sub < 4; ++sub) {
-        bf16x8 kf0 = load_frag_rowmajor(k, kn0 + 16 * sub, 0, lane, HS);
-        bf16x8 kf1 = load_frag_rowmajor(k, kn0 + 16 * sub, 1, lane, HS);
+        bf16x8 kf0 = read_frag_swz(lds_k[buf], 16 * sub, 0, lane);
+        bf16x8 kf1 = read_frag_swz(lds_k[buf], 16 * sub, 1, lane);
         st[sub] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             kf0, qf[0], f32x4{0.f, 0.f, 0.f, 0.f}, 0, 0, 0);
         st[sub] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -875,7 +952,10 @@ __global__ __launch_bounds__(256) void flash_fwd_v3_kernel(
         }
       }
     }
-    // next tile writes the OTHER V^T buffer: no trailing barrier
+    // write-late: next tile's regs -> the other buffers, one barrier
+    stage_t_write(lds_vt[buf ^ 1], vt);
+    stage_r_write(lds_k[buf ^ 1], kt);
+    __syncthreads();
   }
 
   // --- epilogue: transpose O^T -> O rows through LDS ------------------
