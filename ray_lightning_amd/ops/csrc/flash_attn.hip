@@ -1053,10 +1053,10 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_v3_kernel(
   const int mykv = kv0 + (lane & 15);
   const int g = lane >> 4;
 
-  __shared__ __bf16 lds_dot[HS * BN];    // dO^T [hs][q]  8 KB
-  __shared__ __bf16 lds_qt[HS * BN];     // Q^T  [hs][q]  8 KB
-  __shared__ float lds_lse[BN];
-  __shared__ float lds_d[BN];
+  __shared__ __bf16 lds_dot[2][HS * BN]; // dO^T [hs][q] db, 16 KB
+  __shared__ __bf16 lds_qt[2][HS * BN];  // Q^T  [hs][q] db, 16 KB
+  __shared__ float lds_lse[2][BN];
+  __shared__ float lds_d[2][BN];
   __shared__ __bf16 lds_ep[4][WQ * HS];  // epilogue     8 KB
 
   // K^T / V^T B-fragments (col = kv = lane&15): kernel-resident
@@ -1076,14 +1076,27 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_v3_kernel(
 
   const float l2e = 1.4426950408889634f;
 
-  for (int qm0 = kb0; qm0 < T; qm0 += BN) {
-    stage_transposed_swz(lds_dot, dO, qm0);
-    stage_transposed_swz(lds_qt, q, qm0);
+  {  // prologue: first q tile through registers
+    StageRegsT dt = stage_t_load(dO, kb0);
+    StageRegsT qt = stage_t_load(q, kb0);
+    stage_t_write(lds_dot[0], dt);
+    stage_t_write(lds_qt[0], qt);
     if (threadIdx.x < BN) {
-      lds_lse[threadIdx.x] = lse[qm0 + threadIdx.x];
-      lds_d[threadIdx.x] = Drow[qm0 + threadIdx.x];
+      lds_lse[0][threadIdx.x] = lse[kb0 + threadIdx.x];
+      lds_d[0][threadIdx.x] = Drow[kb0 + threadIdx.x];
     }
-    __syncthreads();
+  }
+  __syncthreads();
+  int buf = 0;
+  for (int qm0 = kb0; qm0 < T; qm0 += BN, buf ^= 1) {
+    const int next = (qm0 + BN < T) ? qm0 + BN : qm0;
+    StageRegsT dt = stage_t_load(dO, next);
+    StageRegsT qt = stage_t_load(q, next);
+    float nlse = 0.f, nd = 0.f;
+    if (threadIdx.x < BN) {
+      nlse = lse[next + threadIdx.x];
+      nd = Drow[next + threadIdx.x];
+    }
     const bool active = kv0 <= qm0 + BN - 1;
     if (active) {
       float pv[4][4], dsv[4][4];
@@ -1108,10 +1121,10 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_v3_kernel(
           const int qi = 16 * sub + 4 * g + r;     // tile-local q
           const bool valid = (qm0 + qi) >= mykv;
           const float p = valid
-              ? exp2f((scale * sacc[r] - lds_lse[qi]) * l2e) : 0.f;
+              ? exp2f((scale * sacc[r] - lds_lse[buf][qi]) * l2e) : 0.f;
           pv[sub][r] = p;
           dsv[sub][r] = valid
-              ? scale * p * (dpacc[r] - lds_d[qi]) : 0.f;
+              ? scale * p * (dpacc[r] - lds_d[buf][qi]) : 0.f;
         }
       }
       // redistribute over the q axis -> B-fragments (k = q)
@@ -1148,14 +1161,22 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_v3_kernel(
         }
         #pragma unroll
         for (int n = 0; n < 4; ++n) {
-          bf16x8 adot = read_frag_swz(lds_dot, 16 * n, blk, lane);
-          bf16x8 aqt = read_frag_swz(lds_qt, 16 * n, blk, lane);
+          bf16x8 adot = read_frag_swz(lds_dot[buf], 16 * n, blk,
+                                      lane);
+          bf16x8 aqt = read_frag_swz(lds_qt[buf], 16 * n, blk, lane);
           dv_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               adot, pb, dv_acc[n], 0, 0, 0);
           dk_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               aqt, db, dk_acc[n], 0, 0, 0);
         }
       }
+    }
+    // write-late: next tile -> other buffers, one barrier per tile
+    stage_t_write(lds_dot[buf ^ 1], dt);
+    stage_t_write(lds_qt[buf ^ 1], qt);
+    if (threadIdx.x < BN) {
+      lds_lse[buf ^ 1][threadIdx.x] = nlse;
+      lds_d[buf ^ 1][threadIdx.x] = nd;
     }
     __syncthreads();
   }
@@ -1191,7 +1212,8 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_v3_kernel(
 //   dP^T[kv][q] = mfma(V_frag,  dO^T B-frag)
 //   dQ^T[d][q]  = mfma(K^T A-frag(lds), redist(dS^T) B-frag)
 template <bool USE_PERMLANE>
-__global__ __launch_bounds__(256) void flash_bwd_dq_v3_kernel(
+// min 4 waves/EU: LDS allows 6, keep the allocator under 128 regs
+__global__ __launch_bounds__(256, 4) void flash_bwd_dq_v3_kernel(
     const __hip_bfloat16* __restrict__ Qg,
     const __hip_bfloat16* __restrict__ Kg,
     const __hip_bfloat16* __restrict__ Vg,
@@ -1213,7 +1235,7 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_v3_kernel(
   const int myq = q0 + (lane & 15);
   const int g = lane >> 4;
 
-  __shared__ __bf16 lds_kt[HS * BN];     // K^T [hs][kv]  8 KB
+  __shared__ __bf16 lds_kt[2][HS * BN];  // K^T [hs][kv] db, 16 KB
   __shared__ __bf16 lds_ep[4][WQ * HS];  // epilogue      8 KB
 
   bf16x8 qf[2], dof[2];
@@ -1231,10 +1253,16 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_v3_kernel(
     dq_acc[n] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const float l2e = 1.4426950408889634f;
-  const int kv_end = qm0 + BM3;
-  for (int kn0 = 0; kn0 < kv_end && kn0 < T; kn0 += BN) {
-    stage_transposed_swz(lds_kt, k, kn0);
-    __syncthreads();
+  const int kv_end = (qm0 + BM3 < T) ? qm0 + BM3 : T;
+  {
+    StageRegsT kt = stage_t_load(k, 0);
+    stage_t_write(lds_kt[0], kt);
+  }
+  __syncthreads();
+  int buf = 0;
+  for (int kn0 = 0; kn0 < kv_end; kn0 += BN, buf ^= 1) {
+    const int next = (kn0 + BN < kv_end) ? kn0 + BN : kn0;
+    StageRegsT kt = stage_t_load(k, next);
     const bool active = kn0 <= q0 + WQ - 1;
     if (active) {
       float dsv[4][4];
@@ -1278,12 +1306,13 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_v3_kernel(
         pw[0] = d0; pw[1] = d1; pw[2] = d2; pw[3] = d3;
         #pragma unroll
         for (int n = 0; n < 4; ++n) {
-          bf16x8 akt = read_frag_swz(lds_kt, 16 * n, blk, lane);
+          bf16x8 akt = read_frag_swz(lds_kt[buf], 16 * n, blk, lane);
           dq_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               akt, db, dq_acc[n], 0, 0, 0);
         }
       }
     }
+    stage_t_write(lds_kt[buf ^ 1], kt);
     __syncthreads();
   }
 
