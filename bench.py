@@ -107,6 +107,17 @@ def parse_args() -> argparse.Namespace:
     if args.nccl_debug:
         os.environ.setdefault("NCCL_DEBUG", "INFO")
         os.environ.setdefault("NCCL_DEBUG_SUBSYS", "INIT,GRAPH,TUNING")
+    if is_gpt:
+        # hipBLASLt algo pinning: load the TunableOp results tuned on
+        # MI355X for the GPT-2-XL shapes (scripts/tune_gemms.py wrote
+        # them; read-only here — no tuning cost in the bench).
+        csvp = os.path.join(
+            os.path.dirname(os.path.abspath(__file__)),
+            "ray_lightning_amd", "ops", "tunableop_gpt2xl.csv")
+        if os.path.exists(csvp):
+            os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+            os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "0")
+            os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME", csvp)
     return args
 
 

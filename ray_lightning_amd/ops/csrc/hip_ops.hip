@@ -659,6 +659,7 @@ torch::Tensor perm_probe(torch::Tensor M, torch::Tensor B,
                          long variant);
 std::vector<torch::Tensor> perm_dump(torch::Tensor M, long variant);
 torch::Tensor permlane_swap_probe();
+torch::Tensor tr16_probe(long a, long b, long c);
 
 // fused residual-add + LayerNorm — defined in fused_ln.hip
 std::vector<torch::Tensor> fused_ln_fwd(
@@ -692,6 +693,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("perm_probe", &perm_probe,
         "C-layout -> A-fragment bpermute redistribution probe");
   m.def("perm_dump", &perm_dump, "redistribution element dump");
+  m.def("tr16_probe", &tr16_probe,
+        "ds_read_b64_tr_b16 lane/element map probe");
   m.def("permlane_swap_probe", &permlane_swap_probe,
         "permlane{16,32}_swap lane-exchange semantics probe");
   m.def("scale_inplace", &scale_inplace, "flat *= s");

@@ -280,3 +280,43 @@ torch::Tensor permlane_swap_probe() {
               hipGetErrorString(e));
   return out;
 }
+
+namespace {
+// ds_read_b64_tr_b16 semantics probe: LDS u16[4096] filled with its own
+// index; each lane reads 8 B at addr = a*lane + b*(lane&15) + c*(lane>>4)
+// and we dump the 4 u16 element-indices each lane received.
+__global__ void tr16_probe_kernel(unsigned short* __restrict__ out,
+                                  int a, int b, int c) {
+  __shared__ unsigned short lds[4096];
+  for (int i = threadIdx.x; i < 4096; i += blockDim.x)
+    lds[i] = (unsigned short)i;
+  __syncthreads();
+  if (threadIdx.x >= 64) return;
+  const int lane = threadIdx.x;
+  const int addr = a * lane + b * (lane & 15) + c * (lane >> 4);
+  unsigned r0, r1;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %2\n"
+      "s_waitcnt lgkmcnt(0)"
+      : "=v"(r0), "=v"(r1)
+      : "v"(reinterpret_cast<unsigned long>(
+            reinterpret_cast<char*>(lds) + addr) & 0xffffffffu)
+      : "memory");
+  out[lane * 4 + 0] = (unsigned short)(r0 & 0xffff);
+  out[lane * 4 + 1] = (unsigned short)(r0 >> 16);
+  out[lane * 4 + 2] = (unsigned short)(r1 & 0xffff);
+  out[lane * 4 + 3] = (unsigned short)(r1 >> 16);
+}
+}  // namespace
+
+torch::Tensor tr16_probe(long a, long b, long c) {
+  auto out = torch::zeros({64, 4}, torch::dtype(at::kShort)
+                                       .device(at::kCUDA));
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(tr16_probe_kernel, dim3(1), dim3(256), 0, stream,
+                     reinterpret_cast<unsigned short*>(out.data_ptr()),
+                     (int)a, (int)b, (int)c);
+  hipError_t e = hipGetLastError();
+  TORCH_CHECK(e == hipSuccess, "tr16_probe: ", hipGetErrorString(e));
+  return out;
+}
