@@ -294,18 +294,22 @@ __global__ void tr16_probe_kernel(unsigned short* __restrict__ out,
   if (threadIdx.x >= 64) return;
   const int lane = threadIdx.x;
   const int addr = a * lane + b * (lane & 15) + c * (lane >> 4);
-  unsigned r0, r1;
+  // AS(3) pointer -> 32-bit LDS byte offset for the ds instruction
+  typedef __attribute__((ext_vector_type(2))) unsigned uint2v;
+  const unsigned base =
+      (unsigned)(size_t)(__attribute__((address_space(3))) char*)
+          (void*)lds;
+  uint2v r;
   asm volatile(
-      "ds_read_b64_tr_b16 %0, %2\n"
+      "ds_read_b64_tr_b16 %0, %1\n"
       "s_waitcnt lgkmcnt(0)"
-      : "=v"(r0), "=v"(r1)
-      : "v"(reinterpret_cast<unsigned long>(
-            reinterpret_cast<char*>(lds) + addr) & 0xffffffffu)
+      : "=v"(r)
+      : "v"(base + (unsigned)addr)
       : "memory");
-  out[lane * 4 + 0] = (unsigned short)(r0 & 0xffff);
-  out[lane * 4 + 1] = (unsigned short)(r0 >> 16);
-  out[lane * 4 + 2] = (unsigned short)(r1 & 0xffff);
-  out[lane * 4 + 3] = (unsigned short)(r1 >> 16);
+  out[lane * 4 + 0] = (unsigned short)(r[0] & 0xffff);
+  out[lane * 4 + 1] = (unsigned short)(r[0] >> 16);
+  out[lane * 4 + 2] = (unsigned short)(r[1] & 0xffff);
+  out[lane * 4 + 3] = (unsigned short)(r[1] >> 16);
 }
 }  // namespace
 
