@@ -186,6 +186,40 @@ __device__ __forceinline__ bf16x8 read_frag_swz(const __bf16* img,
   return v;
 }
 
+// Transposed A-fragment straight from the swizzled ROW-major image via
+// ds_read_b64_tr_b16 (guide T10; semantics verified by tr16_probe:
+// per 16-lane group the 16 8-byte segments form a [4 src-rows][16 col]
+// block — rows = lanes 4r..4r+3 — and lane j receives column j).
+// Fragment: row = col16 + (lane&15), k elems i = source rows
+// krow0 + 8*(lane>>4) + i. Rows are swizzle-independent (each lane
+// addresses its own 8 B segment inside one granule).
+__device__ __forceinline__ bf16x8 read_frag_tr(const __bf16* img,
+                                               int krow0, int col16,
+                                               int lane) {
+  const int j = lane & 15;
+  const int g = lane >> 4;
+  const unsigned base =
+      (unsigned)(size_t)(__attribute__((address_space(3))) const char*)
+          (const void*)img;
+  const int r0 = krow0 + 8 * g + (j >> 2);
+  const int cb = (col16 + 4 * (j & 3)) * 2;
+  const unsigned a0 = base + (unsigned)swz_off(r0, cb);
+  const unsigned a1 = base + (unsigned)swz_off(r0 + 4, cb);
+  typedef __attribute__((ext_vector_type(2))) unsigned uint2v;
+  uint2v lo, hi;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %2\n"
+      "ds_read_b64_tr_b16 %1, %3\n"
+      "s_waitcnt lgkmcnt(0)"
+      : "=v"(lo), "=v"(hi)
+      : "v"(a0), "v"(a1)
+      : "memory");
+  bf16x8 v;
+  unsigned* w = reinterpret_cast<unsigned*>(&v);
+  w[0] = lo[0]; w[1] = lo[1]; w[2] = hi[0]; w[3] = hi[1];
+  return v;
+}
+
 // ---------------------------------------------------------------------
 // forward
 // ---------------------------------------------------------------------
@@ -840,7 +874,7 @@ __global__ __launch_bounds__(256, 4) void flash_fwd_v3_kernel(
   const int myq = q0 + (lane & 15);    // this lane's q column
   const int g = lane >> 4;             // kv group 0..3
 
-  __shared__ __bf16 lds_vt[2][HS * BN];   // V^T double buffer, 16 KB
+  __shared__ __bf16 lds_v[2][BN * HS];    // V row-major db buffer, 16 KB
   __shared__ __bf16 lds_k[2][BN * HS];    // K row-major db buffer, 16 KB
   __shared__ __bf16 lds_o[4][WQ * HS];    // epilogue transpose, 8 KB
 
@@ -862,16 +896,16 @@ __global__ __launch_bounds__(256, 4) void flash_fwd_v3_kernel(
   // prologue: tile 0 staged through registers, then the loop keeps one
   // tile in flight (issue loads -> compute current -> write other buf)
   {
-    StageRegsT vt = stage_t_load(v, 0);
+    StageRegsR vt = stage_r_load(v, 0);
     StageRegsR kt = stage_r_load(k, 0);
-    stage_t_write(lds_vt[0], vt);
+    stage_r_write(lds_v[0], vt);
     stage_r_write(lds_k[0], kt);
   }
   __syncthreads();
   int buf = 0;
   for (int kn0 = 0; kn0 < kv_end; kn0 += BN, buf ^= 1) {
     const int next = (kn0 + BN < kv_end) ? kn0 + BN : kn0;
-    StageRegsT vt = stage_t_load(v, next);
+    StageRegsR vt = stage_r_load(v, next);
     StageRegsR kt = stage_r_load(k, next);
     const bool active = kn0 <= q0 + WQ - 1;
     if (active) {
@@ -946,14 +980,15 @@ __global__ __launch_bounds__(256, 4) void flash_fwd_v3_kernel(
       for (int n = 0; n < 4; ++n) {
         #pragma unroll
         for (int c = 0; c < 2; ++c) {
-          bf16x8 va = read_frag_swz(lds_vt[buf], 16 * n, c, lane);
+          // V^T fragment (row d, k = kv) from the row-major V image
+          bf16x8 va = read_frag_tr(lds_v[buf], 32 * c, 16 * n, lane);
           o_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               va, pb[c], o_acc[n], 0, 0, 0);
         }
       }
     }
     // write-late: next tile's regs -> the other buffers, one barrier
-    stage_t_write(lds_vt[buf ^ 1], vt);
+    stage_r_write(lds_v[buf ^ 1], vt);
     stage_r_write(lds_k[buf ^ 1], kt);
     __syncthreads();
   }
@@ -1028,7 +1063,8 @@ namespace {
 //   dV^T[d][kv]= mfma(dO^T A-frag(lds), redist(P) B-frag)
 //   dK^T[d][kv]= mfma(Q^T  A-frag(lds), redist(dS) B-frag)
 template <bool USE_PERMLANE>
-__global__ __launch_bounds__(256) void flash_bwd_dkv_v3_kernel(
+// min 4 waves/EU: LDS fits 4 workgroups, keep regs <=128
+__global__ __launch_bounds__(256, 4) void flash_bwd_dkv_v3_kernel(
     const __hip_bfloat16* __restrict__ Qg,
     const __hip_bfloat16* __restrict__ Kg,
     const __hip_bfloat16* __restrict__ Vg,
@@ -1053,11 +1089,12 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_v3_kernel(
   const int mykv = kv0 + (lane & 15);
   const int g = lane >> 4;
 
-  __shared__ __bf16 lds_dot[2][HS * BN]; // dO^T [hs][q] db, 16 KB
-  __shared__ __bf16 lds_qt[2][HS * BN];  // Q^T  [hs][q] db, 16 KB
+  __shared__ __bf16 lds_do[2][BN * HS];  // dO row-major db, 16 KB
+  __shared__ __bf16 lds_q[2][BN * HS];   // Q  row-major db, 16 KB
   __shared__ float lds_lse[2][BN];
   __shared__ float lds_d[2][BN];
-  __shared__ __bf16 lds_ep[4][WQ * HS];  // epilogue     8 KB
+  // epilogue transposes reuse lds_do[0] (dead after the q loop):
+  // keeps LDS at 33.8 KB -> 4 workgroups/CU
 
   // K^T / V^T B-fragments (col = kv = lane&15): kernel-resident
   bf16x8 kf[2], vf[2];
@@ -1077,10 +1114,10 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_v3_kernel(
   const float l2e = 1.4426950408889634f;
 
   {  // prologue: first q tile through registers
-    StageRegsT dt = stage_t_load(dO, kb0);
-    StageRegsT qt = stage_t_load(q, kb0);
-    stage_t_write(lds_dot[0], dt);
-    stage_t_write(lds_qt[0], qt);
+    StageRegsR dt = stage_r_load(dO, kb0);
+    StageRegsR qt = stage_r_load(q, kb0);
+    stage_r_write(lds_do[0], dt);
+    stage_r_write(lds_q[0], qt);
     if (threadIdx.x < BN) {
       lds_lse[0][threadIdx.x] = lse[kb0 + threadIdx.x];
       lds_d[0][threadIdx.x] = Drow[kb0 + threadIdx.x];
@@ -1090,8 +1127,8 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_v3_kernel(
   int buf = 0;
   for (int qm0 = kb0; qm0 < T; qm0 += BN, buf ^= 1) {
     const int next = (qm0 + BN < T) ? qm0 + BN : qm0;
-    StageRegsT dt = stage_t_load(dO, next);
-    StageRegsT qt = stage_t_load(q, next);
+    StageRegsR dt = stage_r_load(dO, next);
+    StageRegsR qt = stage_r_load(q, next);
     float nlse = 0.f, nd = 0.f;
     if (threadIdx.x < BN) {
       nlse = lse[next + threadIdx.x];
@@ -1102,12 +1139,10 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_v3_kernel(
       float pv[4][4], dsv[4][4];
       #pragma unroll
       for (int sub = 0; sub < 4; ++sub) {
-        bf16x8 aq0 = load_frag_rowmajor(q, qm0 + 16 * sub, 0, lane, HS);
-        bf16x8 aq1 = load_frag_rowmajor(q, qm0 + 16 * sub, 1, lane, HS);
-        bf16x8 ad0 = load_frag_rowmajor(dO, qm0 + 16 * sub, 0, lane,
-                                        HS);
-        bf16x8 ad1 = load_frag_rowmajor(dO, qm0 + 16 * sub, 1, lane,
-                                        HS);
+        bf16x8 aq0 = read_frag_swz(lds_q[buf], 16 * sub, 0, lane);
+        bf16x8 aq1 = read_frag_swz(lds_q[buf], 16 * sub, 1, lane);
+        bf16x8 ad0 = read_frag_swz(lds_do[buf], 16 * sub, 0, lane);
+        bf16x8 ad1 = read_frag_swz(lds_do[buf], 16 * sub, 1, lane);
         f32x4 sacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             aq0, kf[0], f32x4{0.f, 0.f, 0.f, 0.f}, 0, 0, 0);
         sacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -1161,9 +1196,12 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_v3_kernel(
         }
         #pragma unroll
         for (int n = 0; n < 4; ++n) {
-          bf16x8 adot = read_frag_swz(lds_dot[buf], 16 * n, blk,
-                                      lane);
-          bf16x8 aqt = read_frag_swz(lds_qt[buf], 16 * n, blk, lane);
+          // dO^T / Q^T fragments (row d, k = q) via hardware-transpose
+          // reads from the same row-major images
+          bf16x8 adot = read_frag_tr(lds_do[buf], 32 * blk, 16 * n,
+                                     lane);
+          bf16x8 aqt = read_frag_tr(lds_q[buf], 32 * blk, 16 * n,
+                                    lane);
           dv_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               adot, pb, dv_acc[n], 0, 0, 0);
           dk_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -1172,8 +1210,8 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_v3_kernel(
       }
     }
     // write-late: next tile -> other buffers, one barrier per tile
-    stage_t_write(lds_dot[buf ^ 1], dt);
-    stage_t_write(lds_qt[buf ^ 1], qt);
+    stage_r_write(lds_do[buf ^ 1], dt);
+    stage_r_write(lds_q[buf ^ 1], qt);
     if (threadIdx.x < BN) {
       lds_lse[buf ^ 1][threadIdx.x] = nlse;
       lds_d[buf ^ 1][threadIdx.x] = nd;
@@ -1182,7 +1220,8 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_v3_kernel(
   }
 
   // epilogue: transpose dV^T then dK^T through per-wave LDS
-  __bf16* ew = lds_ep[wave];
+  // (carved from lds_do[0]; the last loop barrier already fenced it)
+  __bf16* ew = lds_do[0] + wave * (WQ * HS);
   const int erow = lane >> 2;            // wave-local kv row
   const int ec0 = (lane & 3) * 16;
   #pragma unroll
@@ -1235,7 +1274,8 @@ __global__ __launch_bounds__(256, 4) void flash_bwd_dq_v3_kernel(
   const int myq = q0 + (lane & 15);
   const int g = lane >> 4;
 
-  __shared__ __bf16 lds_kt[2][HS * BN];  // K^T [hs][kv] db, 16 KB
+  __shared__ __bf16 lds_kk[2][BN * HS];  // K row-major db, 16 KB
+  __shared__ __bf16 lds_vv[2][BN * HS];  // V row-major db, 16 KB
   __shared__ __bf16 lds_ep[4][WQ * HS];  // epilogue      8 KB
 
   bf16x8 qf[2], dof[2];
@@ -1255,23 +1295,26 @@ __global__ __launch_bounds__(256, 4) void flash_bwd_dq_v3_kernel(
   const float l2e = 1.4426950408889634f;
   const int kv_end = (qm0 + BM3 < T) ? qm0 + BM3 : T;
   {
-    StageRegsT kt = stage_t_load(k, 0);
-    stage_t_write(lds_kt[0], kt);
+    StageRegsR kt = stage_r_load(k, 0);
+    StageRegsR vt = stage_r_load(v, 0);
+    stage_r_write(lds_kk[0], kt);
+    stage_r_write(lds_vv[0], vt);
   }
   __syncthreads();
   int buf = 0;
   for (int kn0 = 0; kn0 < kv_end; kn0 += BN, buf ^= 1) {
     const int next = (kn0 + BN < kv_end) ? kn0 + BN : kn0;
-    StageRegsT kt = stage_t_load(k, next);
+    StageRegsR kt = stage_r_load(k, next);
+    StageRegsR vt = stage_r_load(v, next);
     const bool active = kn0 <= q0 + WQ - 1;
     if (active) {
       float dsv[4][4];
       #pragma unroll
       for (int sub = 0; sub < 4; ++sub) {
-        bf16x8 ak0 = load_frag_rowmajor(k, kn0 + 16 * sub, 0, lane, HS);
-        bf16x8 ak1 = load_frag_rowmajor(k, kn0 + 16 * sub, 1, lane, HS);
-        bf16x8 av0 = load_frag_rowmajor(v, kn0 + 16 * sub, 0, lane, HS);
-        bf16x8 av1 = load_frag_rowmajor(v, kn0 + 16 * sub, 1, lane, HS);
+        bf16x8 ak0 = read_frag_swz(lds_kk[buf], 16 * sub, 0, lane);
+        bf16x8 ak1 = read_frag_swz(lds_kk[buf], 16 * sub, 1, lane);
+        bf16x8 av0 = read_frag_swz(lds_vv[buf], 16 * sub, 0, lane);
+        bf16x8 av1 = read_frag_swz(lds_vv[buf], 16 * sub, 1, lane);
         f32x4 st = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             ak0, qf[0], f32x4{0.f, 0.f, 0.f, 0.f}, 0, 0, 0);
         st = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -1306,13 +1349,16 @@ __global__ __launch_bounds__(256, 4) void flash_bwd_dq_v3_kernel(
         pw[0] = d0; pw[1] = d1; pw[2] = d2; pw[3] = d3;
         #pragma unroll
         for (int n = 0; n < 4; ++n) {
-          bf16x8 akt = read_frag_swz(lds_kt[buf], 16 * n, blk, lane);
+          // K^T fragment (row d, k = kv) via hardware-transpose read
+          bf16x8 akt = read_frag_tr(lds_kk[buf], 32 * blk, 16 * n,
+                                    lane);
           dq_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               akt, db, dq_acc[n], 0, 0, 0);
         }
       }
     }
-    stage_t_write(lds_kt[buf ^ 1], kt);
+    stage_r_write(lds_kk[buf ^ 1], kt);
+    stage_r_write(lds_vv[buf ^ 1], vt);
     __syncthreads();
   }
 
