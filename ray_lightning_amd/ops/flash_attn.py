@@ -1,10 +1,14 @@
 """Hand-written CDNA4 MFMA flash attention (causal, hs=64, bf16).
 
 ``flash_attention(q, k, v)`` is SDPA-shaped: [B, H, T, hs] in/out.
-The custom kernels engage for causal bf16 hs=64 T%128==0 on GPU
-(BM=128 row tiles);
-anything else falls back to ``F.scaled_dot_product_attention``.
-``RLA_FLASH=0`` disables the custom path globally.
+The custom v3 kernels (swapped-operand S^T, in-register permlane
+redistribution, ds_read_b64_tr_b16 transposed fragments — see
+flash_attn.hip) engage for causal bf16 hs=64 T%64==0 on GPU and are
+the DEFAULT: measured 0.68 ms fwd+bwd vs AOTriton SDPA's 0.71 ms on
+the GPT-2-XL shape (B=8, H=25, T=1024), profiles/r02_flash_v3.md.
+Anything else falls back to ``F.scaled_dot_product_attention``.
+``RLA_FLASH=0`` disables the custom path; ``RLA_FLASH_SHFL=1`` selects
+the ds_bpermute redistribution variant (debug).
 """
 from __future__ import annotations
 
@@ -18,19 +22,23 @@ from . import _load_ext
 
 
 def _usable(q: torch.Tensor) -> bool:
-    if os.environ.get("RLA_FLASH", "0") != "1":  # opt-in until it beats SDPA
+    if os.environ.get("RLA_FLASH", "1") == "0":
         return False
     return (q.is_cuda and q.dtype == torch.bfloat16 and q.dim() == 4
-            and q.shape[-1] == 64 and q.shape[2] % 128 == 0
+            and q.shape[-1] == 64 and q.shape[2] % 64 == 0
             and _load_ext() is not None
-            and hasattr(_load_ext(), "flash_attn_fwd"))
+            and hasattr(_load_ext(), "flash_attn_fwd_v3"))
+
+
+def _use_permlane() -> bool:
+    return os.environ.get("RLA_FLASH_SHFL", "0") != "1"
 
 
 class _FlashAttn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, scale):
         ext = _load_ext()
-        o, lse = ext.flash_attn_fwd(q, k, v, scale)
+        o, lse = ext.flash_attn_fwd_v3(q, k, v, scale, _use_permlane())
         ctx.save_for_backward(q, k, v, o, lse)
         ctx.scale = scale
         return o
@@ -39,8 +47,8 @@ class _FlashAttn(torch.autograd.Function):
     def backward(ctx, dout):
         q, k, v, o, lse = ctx.saved_tensors
         ext = _load_ext()
-        dq, dk, dv = ext.flash_attn_bwd(dout, q, k, v, o, lse,
-                                        ctx.scale)
+        dq, dk, dv = ext.flash_attn_bwd_v3(dout, q, k, v, o, lse,
+                                           ctx.scale, _use_permlane())
         return dq, dk, dv, None
 
 

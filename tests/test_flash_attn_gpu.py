@@ -125,7 +125,9 @@ def test_flash_vs_sdpa_speed():
         q, k, v, is_causal=True))
     print(f"\n[flash] custom {t_custom:.3f} ms  sdpa {t_sdpa:.3f} ms "
           f"(fwd+bwd, B{B} H{H} T{T})")
-    assert t_custom < t_sdpa * 3  # sanity: not catastrophically slow
+    # v3 must stay at least on par with AOTriton (it wins by ~4% as of
+    # r02: 0.68 vs 0.71 ms); 15% headroom for box-to-box variance
+    assert t_custom < t_sdpa * 1.15
 
 
 @pytest.mark.xfail(reason="single-bpermute-per-pair scheme is "
