@@ -220,6 +220,44 @@ __device__ __forceinline__ bf16x8 read_frag_tr(const __bf16* img,
   return v;
 }
 
+// Paired variant: transposed fragments from TWO images with a single
+// lgkm drain (dkv reads dO^T and Q^T back to back — separate helper
+// calls would serialize on their internal waits).
+__device__ __forceinline__ void read_frag_tr2(const __bf16* imgA,
+                                              const __bf16* imgB,
+                                              int krow0, int col16,
+                                              int lane, bf16x8& va,
+                                              bf16x8& vb) {
+  const int j = lane & 15;
+  const int g = lane >> 4;
+  const unsigned baseA =
+      (unsigned)(size_t)(__attribute__((address_space(3))) const char*)
+          (const void*)imgA;
+  const unsigned baseB =
+      (unsigned)(size_t)(__attribute__((address_space(3))) const char*)
+          (const void*)imgB;
+  const int r0 = krow0 + 8 * g + (j >> 2);
+  const int cb = (col16 + 4 * (j & 3)) * 2;
+  const unsigned o0 = (unsigned)swz_off(r0, cb);
+  const unsigned o1 = (unsigned)swz_off(r0 + 4, cb);
+  typedef __attribute__((ext_vector_type(2))) unsigned uint2v;
+  uint2v a0, a1, b0, b1;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %4\n"
+      "ds_read_b64_tr_b16 %1, %5\n"
+      "ds_read_b64_tr_b16 %2, %6\n"
+      "ds_read_b64_tr_b16 %3, %7\n"
+      "s_waitcnt lgkmcnt(0)"
+      : "=v"(a0), "=v"(a1), "=v"(b0), "=v"(b1)
+      : "v"(baseA + o0), "v"(baseA + o1), "v"(baseB + o0),
+        "v"(baseB + o1)
+      : "memory");
+  unsigned* wa = reinterpret_cast<unsigned*>(&va);
+  wa[0] = a0[0]; wa[1] = a0[1]; wa[2] = a1[0]; wa[3] = a1[1];
+  unsigned* wb = reinterpret_cast<unsigned*>(&vb);
+  wb[0] = b0[0]; wb[1] = b0[1]; wb[2] = b1[0]; wb[3] = b1[1];
+}
+
 // ---------------------------------------------------------------------
 // forward
 // ---------------------------------------------------------------------
@@ -938,8 +976,25 @@ __global__ __launch_bounds__(256, 4) void flash_fwd_v3_kernel(
           mx = fmaxf(mx, st[sub][r]);
       mx = fmaxf(mx, __shfl_xor(mx, 16, 64));
       mx = fmaxf(mx, __shfl_xor(mx, 32, 64));
-      const float m_new = fmaxf(m_i, mx);
-      const float corr = exp2f((m_i - m_new) * l2e);
+      // T13 defer-max: if no lane's max grew by more than 8 exponent
+      // units, keep the old running max and skip the O rescale —
+      // P is then bounded by 2^8, which the f32 accumulators absorb.
+      // Decision covers THIS tile's P before it is exponentiated
+      // (textbook-safe order); first tile never defers (m_i = -1e30).
+      const bool defer = __all((mx - m_i) * l2e <= 8.0f);
+      float m_new, corr;
+      if (defer) {
+        m_new = m_i;
+        corr = 1.0f;
+      } else {
+        m_new = fmaxf(m_i, mx);
+        corr = exp2f((m_i - m_new) * l2e);
+        #pragma unroll
+        for (int n = 0; n < 4; ++n)
+          #pragma unroll
+          for (int r = 0; r < 4; ++r)
+            o_acc[n][r] *= corr;
+      }
       float rowsum = 0.f;
       #pragma unroll
       for (int sub = 0; sub < 4; ++sub)
@@ -954,11 +1009,6 @@ __global__ __launch_bounds__(256, 4) void flash_fwd_v3_kernel(
       rowsum += __shfl_xor(rowsum, 32, 64);
       l_i = l_i * corr + rowsum;
       m_i = m_new;
-      #pragma unroll
-      for (int n = 0; n < 4; ++n)
-        #pragma unroll
-        for (int r = 0; r < 4; ++r)
-          o_acc[n][r] *= corr;
       // --- P^T -> PV B-fragments, in registers -----------------------
       bf16x8 pb[2];
       #pragma unroll
@@ -1197,11 +1247,10 @@ __global__ __launch_bounds__(256, 4) void flash_bwd_dkv_v3_kernel(
         #pragma unroll
         for (int n = 0; n < 4; ++n) {
           // dO^T / Q^T fragments (row d, k = q) via hardware-transpose
-          // reads from the same row-major images
-          bf16x8 adot = read_frag_tr(lds_do[buf], 32 * blk, 16 * n,
-                                     lane);
-          bf16x8 aqt = read_frag_tr(lds_q[buf], 32 * blk, 16 * n,
-                                    lane);
+          // reads from the same row-major images (one lgkm drain)
+          bf16x8 adot, aqt;
+          read_frag_tr2(lds_do[buf], lds_q[buf], 32 * blk, 16 * n,
+                        lane, adot, aqt);
           dv_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               adot, pb, dv_acc[n], 0, 0, 0);
           dk_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(

@@ -225,3 +225,25 @@ def test_flash_bwd_v3_numerics(B, H, T, use_permlane):
                             (dv, vf.grad, "dv")):
         err = (got.float() - want).abs().max().item()
         assert err < 6e-3 * math.sqrt(T / 64), f"{name} max err {err}"
+
+
+def test_flash_fwd_v3_defer_max_spike():
+    """T13 hazard test (guide §5.4 rule 26): an input engineered to
+    FORCE the defer-rescale branch (a late spiked key that jumps the
+    running max far beyond the threshold) must match the fp32
+    reference exactly as well as random data does."""
+    from ray_lightning_amd.ops import _load_ext
+    ext = _load_ext()
+    torch.manual_seed(11)
+    B, H, T, hs = 1, 2, 512, 64
+    q = (torch.randn(B, H, T, hs, device="cuda") * 0.3).bfloat16()
+    k = (torch.randn(B, H, T, hs, device="cuda") * 0.3).bfloat16()
+    v = (torch.randn(B, H, T, hs, device="cuda") * 0.3).bfloat16()
+    # spike: key row 400 aligned with every query -> at its tile the
+    # raw score jumps by >> 8/log2(e)*scale exponent units
+    k[:, :, 400, :] = (q[:, :, -1, :] * 40.0).bfloat16()
+    scale = 1.0 / math.sqrt(hs)
+    o, lse = ext.flash_attn_fwd_v3(q, k, v, scale, True)
+    ref = _ref_attention(q, k, v, scale)
+    err = (o.float() - ref).abs().max().item()
+    assert err < 3e-2, f"defer-max spike: max err {err}"
