@@ -30,7 +30,7 @@ def test_gpt2_tiny_trains_on_fused_stack():
     torch.manual_seed(0)
     cfg = _tiny_cfg()
     model = to_bf16_training(GPT2(cfg).cuda())
-    from ray_lightning_amd.optim import FusedAdam
+    from ray_lightning_amd.optim import FusedAdamW as FusedAdam
     opt = FusedAdam(model.parameters(), lr=1e-3)
     x = torch.randint(0, cfg.vocab_size, (4, 128), device="cuda")
     y = torch.randint(0, cfg.vocab_size, (4, 128), device="cuda")
