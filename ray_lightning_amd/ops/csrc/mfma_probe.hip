@@ -324,3 +324,46 @@ torch::Tensor tr16_probe(long a, long b, long c) {
   TORCH_CHECK(e == hipSuccess, "tr16_probe: ", hipGetErrorString(e));
   return out;
 }
+
+namespace {
+using f32x16 = __attribute__((ext_vector_type(16))) float;
+
+// 32x32x16 bf16 layout probe. Assumed maps (guide §3 + CDNA pattern):
+//   A[32M x 16K]: lane l elem i -> row = l & 31, k = (l >> 5) * 8 + i
+//   B[16K x 32N]: lane l elem i -> col = l & 31, k = (l >> 5) * 8 + i
+//   C (16 f32):   col = l & 31, row = (r & 3) + 8 * (r >> 2) + 4 * (l >> 5)
+__global__ void mfma_probe_32x32x16(const __hip_bfloat16* __restrict__ A,
+                                    const __hip_bfloat16* __restrict__ B,
+                                    float* __restrict__ C) {
+  const int l = threadIdx.x;
+  if (l >= 64) return;
+  bf16x8 a, b;
+  #pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    a[i] = (__bf16)(float)A[(l & 31) * 16 + (l >> 5) * 8 + i];
+    b[i] = (__bf16)(float)B[((l >> 5) * 8 + i) * 32 + (l & 31)];
+  }
+  f32x16 acc = {};
+  acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc, 0, 0, 0);
+  #pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int row = (r & 3) + 8 * (r >> 2) + 4 * (l >> 5);
+    C[row * 32 + (l & 31)] = acc[r];
+  }
+}
+}  // namespace
+
+torch::Tensor mfma_probe32(torch::Tensor A, torch::Tensor B) {
+  TORCH_CHECK(A.sizes() == at::IntArrayRef({32, 16}) &&
+              B.sizes() == at::IntArrayRef({16, 32}));
+  auto C = torch::zeros({32, 32},
+                        A.options().dtype(at::kFloat));
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(mfma_probe_32x32x16, dim3(1), dim3(64), 0, stream,
+      reinterpret_cast<const __hip_bfloat16*>(A.data_ptr()),
+      reinterpret_cast<const __hip_bfloat16*>(B.data_ptr()),
+      C.data_ptr<float>());
+  hipError_t e = hipGetLastError();
+  TORCH_CHECK(e == hipSuccess, "mfma_probe32: ", hipGetErrorString(e));
+  return C;
+}

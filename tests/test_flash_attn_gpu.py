@@ -247,3 +247,17 @@ def test_flash_fwd_v3_defer_max_spike():
     ref = _ref_attention(q, k, v, scale)
     err = (o.float() - ref).abs().max().item()
     assert err < 3e-2, f"defer-max spike: max err {err}"
+
+
+def test_mfma_probe32_layout():
+    """Verifies the assumed 32x32x16 bf16 fragment maps against
+    torch.matmul with asymmetric operands (guide A=I-check rule)."""
+    from ray_lightning_amd import ops
+    ext = ops._load_ext()
+    torch.manual_seed(5)
+    A = (torch.randn(32, 16, device="cuda") * 0.5).bfloat16()
+    B = (torch.randn(16, 32, device="cuda") * 0.5).bfloat16()
+    C = ext.mfma_probe32(A, B)
+    ref = A.float() @ B.float()
+    assert torch.allclose(C, ref, atol=3e-2, rtol=3e-2), \
+        f"32x32x16 layout mismatch: max err {(C - ref).abs().max()}"
