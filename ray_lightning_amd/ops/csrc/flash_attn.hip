@@ -1475,3 +1475,271 @@ std::vector<torch::Tensor> flash_attn_bwd_v3(
   TORCH_CHECK(e == hipSuccess, "flash_bwd_v3: ", hipGetErrorString(e));
   return {dq, dk, dv};
 }
+
+// =====================================================================
+// forward v4 — 32x32x16 MFMA shape: half the MFMA instructions per
+// query of v3 (fwd v3 is issue-bound: SQ_WAIT_INST 42%, MFMA-busy 19%).
+// Same swapped-operand structure; P^T -> PV B-fragments need only ONE
+// permlane32_swap per dword pair (lanes l and l+32 share the q column
+// in BOTH the S^T C-layout and the PV B-layout).
+// Layouts (mfma_probe32-verified):
+//   A[32Mx16K]: lane l elem i -> row l&31, k (l>>5)*8+i
+//   B[16Kx32N]: lane l elem i -> col l&31, k (l>>5)*8+i
+//   C (16 f32): col l&31, row (r&3) + 8*(r>>2) + 4*(l>>5)
+// =====================================================================
+namespace {
+
+using f32x16 = __attribute__((ext_vector_type(16))) float;
+
+constexpr int WQ4 = 32;    // q columns per wave
+constexpr int BM4 = 128;   // q rows per workgroup (4 waves)
+
+// b128 A-fragment from the swizzled row-major image, 32-row block:
+// row = row0 + (lane&31), k-chunk c of 16: byte = c*32 + (lane>>5)*16
+__device__ __forceinline__ bf16x8 read_frag_swz32(const __bf16* img,
+                                                  int row0, int c,
+                                                  int lane) {
+  const int row = row0 + (lane & 31);
+  bf16x8 v;
+  *reinterpret_cast<int4*>(&v) = *reinterpret_cast<const int4*>(
+      reinterpret_cast<const char*>(img) +
+      swz_off(row, 32 * c + 16 * (lane >> 5)));
+  return v;
+}
+
+// Transposed A-fragment via ds_read_b64_tr_b16 with explicit row/col
+// bases (tr16_probe semantics: 16-lane group -> [4 src-rows][16 col]
+// block, lane j of the group receives column j).
+__device__ __forceinline__ bf16x8 read_frag_tr4(const __bf16* img,
+                                                int krow0, int col16,
+                                                int lane) {
+  const int j = lane & 15;
+  const unsigned base =
+      (unsigned)(size_t)(__attribute__((address_space(3))) const char*)
+          (const void*)img;
+  const int r0 = krow0 + (j >> 2);
+  const int cb = (col16 + 4 * (j & 3)) * 2;
+  const unsigned a0 = base + (unsigned)swz_off(r0, cb);
+  const unsigned a1 = base + (unsigned)swz_off(r0 + 4, cb);
+  typedef __attribute__((ext_vector_type(2))) unsigned uint2v;
+  uint2v lo, hi;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %2\n"
+      "ds_read_b64_tr_b16 %1, %3\n"
+      "s_waitcnt lgkmcnt(0)"
+      : "=v"(lo), "=v"(hi)
+      : "v"(a0), "v"(a1)
+      : "memory");
+  bf16x8 v;
+  unsigned* w = reinterpret_cast<unsigned*>(&v);
+  w[0] = lo[0]; w[1] = lo[1]; w[2] = hi[0]; w[3] = hi[1];
+  return v;
+}
+
+__global__ __launch_bounds__(256, 4) void flash_fwd_v4_kernel(
+    const __hip_bfloat16* __restrict__ Q,
+    const __hip_bfloat16* __restrict__ K,
+    const __hip_bfloat16* __restrict__ V,
+    __hip_bfloat16* __restrict__ O, float* __restrict__ LSE,
+    int T, float scale) {
+  const int bh = blockIdx.y;
+  const int qm0 = blockIdx.x * BM4;
+  if (qm0 >= T) return;
+  const long base = (long)bh * T * HS;
+  const __hip_bfloat16* q = Q + base;
+  const __hip_bfloat16* k = K + base;
+  const __hip_bfloat16* v = V + base;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int q0 = qm0 + wave * WQ4;
+  const int myq = q0 + (lane & 31);
+  const int hi = lane >> 5;
+
+  __shared__ __bf16 lds_v[2][BN * HS];   // V row-major db, 16 KB
+  __shared__ __bf16 lds_k[2][BN * HS];   // K row-major db, 16 KB
+
+  // Q^T B-fragments: col = q = lane&31, k = hs = 4 chunks of 16
+  bf16x8 qf[4];
+  #pragma unroll
+  for (int c = 0; c < 4; ++c) {
+    *reinterpret_cast<int4*>(&qf[c]) =
+        *reinterpret_cast<const int4*>(
+            q + (long)(q0 + (lane & 31)) * HS + c * 16 + hi * 8);
+  }
+
+  float m_i = kNegInf, l_i = 0.f;
+  f32x16 o_acc[2];   // O^T: row d = 32*dt + (r&3)+8*(r>>2)+4*hi
+  o_acc[0] = f32x16{};
+  o_acc[1] = f32x16{};
+
+  const float l2e = 1.4426950408889634f * scale;
+  const int kv_end = (qm0 + BM4 < T) ? qm0 + BM4 : T;
+  {
+    StageRegsR vt = stage_r_load(v, 0);
+    StageRegsR kt = stage_r_load(k, 0);
+    stage_r_write(lds_v[0], vt);
+    stage_r_write(lds_k[0], kt);
+  }
+  __syncthreads();
+  int buf = 0;
+  for (int kn0 = 0; kn0 < kv_end; kn0 += BN, buf ^= 1) {
+    const int next = (kn0 + BN < kv_end) ? kn0 + BN : kn0;
+    StageRegsR vt = stage_r_load(v, next);
+    StageRegsR kt = stage_r_load(k, next);
+    const bool active = kn0 <= q0 + WQ4 - 1;
+    if (active) {
+      // --- S^T = K Q^T: lane holds 32 kv of column myq --------------
+      f32x16 st[2];
+      #pragma unroll
+      for (int sub = 0; sub < 2; ++sub) {
+        st[sub] = f32x16{};
+        #pragma unroll
+        for (int c = 0; c < 4; ++c) {
+          bf16x8 kf = read_frag_swz32(lds_k[buf], 32 * sub, c, lane);
+          st[sub] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              kf, qf[c], st[sub], 0, 0, 0);
+        }
+      }
+      // --- causal mask ----------------------------------------------
+      if (kn0 + BN - 1 > q0) {
+        #pragma unroll
+        for (int sub = 0; sub < 2; ++sub)
+          #pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            const int kv = kn0 + 32 * sub + (r & 3) + 8 * (r >> 2) +
+                           4 * hi;
+            if (kv > myq) st[sub][r] = kNegInf;
+          }
+      }
+      // --- online softmax: lane-local over 32 + ONE cross-lane ------
+      float mx = kNegInf;
+      #pragma unroll
+      for (int sub = 0; sub < 2; ++sub)
+        #pragma unroll
+        for (int r = 0; r < 16; ++r)
+          mx = fmaxf(mx, st[sub][r]);
+      mx = fmaxf(mx, __shfl_xor(mx, 32, 64));
+      const bool defer = __all((mx - m_i) * l2e <= 8.0f);
+      float m_new, corr;
+      if (defer) {
+        m_new = m_i;
+        corr = 1.0f;
+      } else {
+        m_new = fmaxf(m_i, mx);
+        corr = exp2f((m_i - m_new) * l2e);
+        #pragma unroll
+        for (int dt = 0; dt < 2; ++dt)
+          #pragma unroll
+          for (int r = 0; r < 16; ++r)
+            o_acc[dt][r] *= corr;
+      }
+      float rowsum = 0.f;
+      #pragma unroll
+      for (int sub = 0; sub < 2; ++sub)
+        #pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const float p = (st[sub][r] <= kNegInf * 0.5f)
+              ? 0.f : exp2f((st[sub][r] - m_new) * l2e);
+          st[sub][r] = p;
+          rowsum += p;
+        }
+      rowsum += __shfl_xor(rowsum, 32, 64);
+      l_i = l_i * corr + rowsum;
+      m_i = m_new;
+      // --- P^T -> PV B-fragments: one pl32swap per dword pair -------
+      // chunk t (kv 16t..16t+15): lane(hi) needs kv 16t+8*hi+i.
+      // From C regs of sub=t>>1: rows 16*(t&1)+{0..3}+4hi (regs
+      // 8*(t&1)+0..3) and 16*(t&1)+{8..11}+4hi (regs 8*(t&1)+4..7):
+      //   L_j = pack(rows 4j block), H_j = pack(rows 8+4j block)
+      //   (P_j, S_j) = pl32swap(L_j, H_j) -> frag = [P0, P1, S0, S1]
+      bf16x8 pb[4];
+      #pragma unroll
+      for (int t = 0; t < 4; ++t) {
+        const float* sv = reinterpret_cast<const float*>(&st[t >> 1]);
+        const int rb = 8 * (t & 1);
+        const unsigned l0 = pack_bf16(sv[rb + 0], sv[rb + 1]);
+        const unsigned l1 = pack_bf16(sv[rb + 2], sv[rb + 3]);
+        const unsigned h0 = pack_bf16(sv[rb + 4], sv[rb + 5]);
+        const unsigned h1 = pack_bf16(sv[rb + 6], sv[rb + 7]);
+        auto ps0 = __builtin_amdgcn_permlane32_swap(l0, h0, false,
+                                                    false);
+        auto ps1 = __builtin_amdgcn_permlane32_swap(l1, h1, false,
+                                                    false);
+        unsigned* w = reinterpret_cast<unsigned*>(&pb[t]);
+        w[0] = ps0[0]; w[1] = ps1[0]; w[2] = ps0[1]; w[3] = ps1[1];
+      }
+      // --- O^T += V^T P^T -------------------------------------------
+      const int g = lane >> 4;
+      #pragma unroll
+      for (int dt = 0; dt < 2; ++dt) {
+        #pragma unroll
+        for (int t = 0; t < 4; ++t) {
+          bf16x8 va = read_frag_tr4(
+              lds_v[buf], 16 * t + 8 * (g >> 1),
+              32 * dt + 16 * (g & 1), lane);
+          o_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              va, pb[t], o_acc[dt], 0, 0, 0);
+        }
+      }
+    }
+    stage_r_write(lds_v[buf ^ 1], vt);
+    stage_r_write(lds_k[buf ^ 1], kt);
+    __syncthreads();
+  }
+
+  // --- epilogue: transpose O^T -> O rows via dead K image LDS -------
+  __bf16* ow = lds_k[0] + wave * (WQ4 * HS);
+  const float inv_l = (l_i > 0.f) ? 1.f / l_i : 0.f;
+  #pragma unroll
+  for (int dt = 0; dt < 2; ++dt)
+    #pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int d = 32 * dt + (r & 3) + 8 * (r >> 2) + 4 * hi;
+      ow[(lane & 31) * HS + d] = (__bf16)(o_acc[dt][r] * inv_l);
+    }
+  wait_lds();
+  {
+    const int row = lane >> 1;           // wave-local q row 0..31
+    const int c0 = (lane & 1) * 32;
+    int4 t0 = *reinterpret_cast<const int4*>(ow + row * HS + c0);
+    int4 t1 = *reinterpret_cast<const int4*>(ow + row * HS + c0 + 8);
+    int4 t2 = *reinterpret_cast<const int4*>(ow + row * HS + c0 + 16);
+    int4 t3 = *reinterpret_cast<const int4*>(ow + row * HS + c0 + 24);
+    __hip_bfloat16* op = O + base + (long)(q0 + row) * HS + c0;
+    *reinterpret_cast<int4*>(op) = t0;
+    *reinterpret_cast<int4*>(op + 8) = t1;
+    *reinterpret_cast<int4*>(op + 16) = t2;
+    *reinterpret_cast<int4*>(op + 24) = t3;
+  }
+  if (hi == 0)
+    LSE[(long)bh * T + myq] = m_i * scale + logf(fmaxf(l_i, 1e-30f));
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> flash_attn_fwd_v4(torch::Tensor q,
+                                             torch::Tensor k,
+                                             torch::Tensor v,
+                                             double scale) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16 &&
+              q.is_contiguous() && k.is_contiguous() &&
+              v.is_contiguous(), "flash_attn_fwd_v4: bf16 contiguous");
+  TORCH_CHECK(q.dim() == 4 && q.size(3) == HS,
+              "flash_attn_fwd_v4: [B,H,T,64] expected");
+  const int B = (int)q.size(0), H = (int)q.size(1), T = (int)q.size(2);
+  TORCH_CHECK(T % BM4 == 0, "flash_attn_fwd_v4: T % 128 == 0");
+  auto o = torch::empty_like(q);
+  auto lse = torch::empty({B, H, T}, q.options().dtype(at::kFloat));
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  dim3 grid(T / BM4, B * H);
+  hipLaunchKernelGGL(flash_fwd_v4_kernel, grid, dim3(256), 0, stream,
+                     reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(k.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(v.data_ptr()),
+                     reinterpret_cast<__hip_bfloat16*>(o.data_ptr()),
+                     lse.data_ptr<float>(), T, (float)scale);
+  hipError_t e = hipGetLastError();
+  TORCH_CHECK(e == hipSuccess, "flash_fwd_v4: ", hipGetErrorString(e));
+  return {o, lse};
+}

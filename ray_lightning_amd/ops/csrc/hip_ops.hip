@@ -661,6 +661,10 @@ std::vector<torch::Tensor> perm_dump(torch::Tensor M, long variant);
 torch::Tensor permlane_swap_probe();
 torch::Tensor tr16_probe(long a, long b, long c);
 torch::Tensor mfma_probe32(torch::Tensor A, torch::Tensor B);
+std::vector<torch::Tensor> flash_attn_fwd_v4(torch::Tensor q,
+                                             torch::Tensor k,
+                                             torch::Tensor v,
+                                             double scale);
 
 // fused residual-add + LayerNorm — defined in fused_ln.hip
 std::vector<torch::Tensor> fused_ln_fwd(
@@ -694,6 +698,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("perm_probe", &perm_probe,
         "C-layout -> A-fragment bpermute redistribution probe");
   m.def("perm_dump", &perm_dump, "redistribution element dump");
+  m.def("flash_attn_fwd_v4", &flash_attn_fwd_v4,
+        "32x32x16 MFMA flash attention fwd (v4)");
   m.def("mfma_probe32", &mfma_probe32,
         "32x32x16 bf16 MFMA layout probe");
   m.def("tr16_probe", &tr16_probe,

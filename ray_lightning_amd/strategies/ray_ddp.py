@@ -151,6 +151,10 @@ class RayStrategy(Strategy):
             return self._device
         if self._external_mode and self.use_gpu:
             local = int(os.environ.get("LOCAL_RANK", "0"))
+            # modulo lets N ranks share fewer devices (gloo-data-plane
+            # validation on small boxes); an 8-GPU deployment maps 1:1
+            if torch.cuda.is_available():
+                local %= max(1, torch.cuda.device_count())
             return torch.device("cuda", local)
         if self._is_remote and self.use_gpu and torch.cuda.is_available():
             gpu_ids = get_gpu_ids()
