@@ -34,7 +34,8 @@ o, lse = ext.flash_attn_fwd(q, k, v, scale)
 t_fwd = t(lambda: ext.flash_attn_fwd(q, k, v, scale))
 t_fwd3s = t(lambda: ext.flash_attn_fwd_v3(q, k, v, scale, False))
 t_fwd3p = t(lambda: ext.flash_attn_fwd_v3(q, k, v, scale, True))
-print(f"fwd v3(shfl) {t_fwd3s:.3f} ms   fwd v3(permlane) {t_fwd3p:.3f} ms")
+t_fwd4 = t(lambda: ext.flash_attn_fwd_v4(q, k, v, scale))
+print(f"fwd v3(shfl) {t_fwd3s:.3f} ms   fwd v3(permlane) {t_fwd3p:.3f} ms   fwd v4(32x32) {t_fwd4:.3f} ms")
 t_bwd = t(lambda: ext.flash_attn_bwd(dy, q, k, v, o, lse, scale))
 t_bwd3s = t(lambda: ext.flash_attn_bwd_v3(dy, q, k, v, o, lse, scale, False))
 t_bwd3p = t(lambda: ext.flash_attn_bwd_v3(dy, q, k, v, o, lse, scale, True))

@@ -261,3 +261,28 @@ def test_mfma_probe32_layout():
     ref = A.float() @ B.float()
     assert torch.allclose(C, ref, atol=3e-2, rtol=3e-2), \
         f"32x32x16 layout mismatch: max err {(C - ref).abs().max()}"
+
+
+@pytest.mark.parametrize("B,H,T", [(2, 3, 256), (1, 2, 1024),
+                                   (1, 1, 128)])
+def test_flash_fwd_v4_numerics(B, H, T):
+    """v4 (32x32x16 MFMA shape) against the fp32 reference + LSE."""
+    from ray_lightning_amd import ops
+    ext = ops._load_ext()
+    torch.manual_seed(4)
+    hs = 64
+    q = (torch.randn(B, H, T, hs, device="cuda") * 0.5).bfloat16()
+    k = (torch.randn(B, H, T, hs, device="cuda") * 0.5).bfloat16()
+    v = (torch.randn(B, H, T, hs, device="cuda") * 0.5).bfloat16()
+    scale = 1.0 / math.sqrt(hs)
+    o, lse = ext.flash_attn_fwd_v4(q, k, v, scale)
+    ref = _ref_attention(q, k, v, scale)
+    assert torch.allclose(o.float(), ref, atol=3e-2, rtol=3e-2), \
+        f"max err {(o.float() - ref).abs().max()}"
+    s = (q.float() @ k.float().transpose(-1, -2)) * scale
+    mask = torch.tril(
+        torch.ones(T, T, device=q.device, dtype=torch.bool))
+    s = s.masked_fill(~mask, float("-inf"))
+    lse_ref = torch.logsumexp(s, dim=-1)
+    assert torch.allclose(lse, lse_ref, atol=2e-3, rtol=1e-3), \
+        f"LSE max err {(lse - lse_ref).abs().max()}"
