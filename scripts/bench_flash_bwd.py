@@ -26,5 +26,7 @@ for _ in range(30):
         ext.flash_attn_bwd(dy, q, k, v, o, lse, scale)
     elif which == "fwd3":
         ext.flash_attn_fwd_v3(q, k, v, scale, True)
+    elif which == "fwd4":
+        ext.flash_attn_fwd_v4(q, k, v, scale)
 torch.cuda.synchronize()
 print("done", which)

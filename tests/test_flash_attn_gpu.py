@@ -224,7 +224,10 @@ def test_flash_bwd_v3_numerics(B, H, T, use_permlane):
     for got, want, name in ((dq, qf.grad, "dq"), (dk, kf.grad, "dk"),
                             (dv, vf.grad, "dv")):
         err = (got.float() - want).abs().max().item()
-        assert err < 6e-3 * math.sqrt(T / 64), f"{name} max err {err}"
+        # fwd runs T13 defer-max (P bounded by 2^8 instead of 1), which
+        # costs ~3x max-abs error in O/D per the guide's measurement —
+        # the bwd inherits it through D = rowsum(dO*O)
+        assert err < 2e-2 * math.sqrt(T / 64), f"{name} max err {err}"
 
 
 def test_flash_fwd_v3_defer_max_spike():
@@ -286,3 +289,26 @@ def test_flash_fwd_v4_numerics(B, H, T):
     lse_ref = torch.logsumexp(s, dim=-1)
     assert torch.allclose(lse, lse_ref, atol=2e-3, rtol=1e-3), \
         f"LSE max err {(lse - lse_ref).abs().max()}"
+
+
+def test_flash_v3_bitwise_deterministic():
+    """Two identical fwd+bwd invocations must agree bitwise — guards
+    against LDS double-buffer / tr_read races that would show up as
+    run-to-run jitter rather than systematic error."""
+    from ray_lightning_amd.ops import _load_ext
+    ext = _load_ext()
+    torch.manual_seed(9)
+    q = torch.randn(2, 3, 256, 64, device="cuda").bfloat16()
+    k = torch.randn_like(q)
+    v = torch.randn_like(q)
+    dy = torch.randn_like(q)
+    outs = []
+    for _ in range(3):
+        o, lse = ext.flash_attn_fwd_v3(q, k, v, 0.125, True)
+        dq, dk, dv = ext.flash_attn_bwd_v3(dy, q, k, v, o, lse, 0.125,
+                                           True)
+        outs.append((o, lse, dq, dk, dv))
+    for a, b in zip(outs[0], outs[1]):
+        assert torch.equal(a, b)
+    for a, b in zip(outs[0], outs[2]):
+        assert torch.equal(a, b)
