@@ -1,4 +1,4 @@
-"""Localize fwd v3 non-determinism: where do repeated runs differ?"""
+"""Localize the fwd/bwd interleave non-determinism: input corruption?"""
 import os
 import sys
 
@@ -13,35 +13,31 @@ B, H, T = 2, 3, 256
 q = torch.randn(B, H, T, 64, device="cuda").bfloat16()
 k = torch.randn_like(q)
 v = torch.randn_like(q)
+dy = torch.randn_like(q)
+qc, kc, vc, dyc = q.clone(), k.clone(), v.clone(), dy.clone()
 
-for pl in (True, False):
-    o0, lse0 = ext.flash_attn_fwd_v3(q, k, v, 0.125, pl)
-    diffs = 0
-    pos = None
-    for i in range(20):
-        o, lse = ext.flash_attn_fwd_v3(q, k, v, 0.125, pl)
-        if not torch.equal(o, o0):
-            diffs += 1
-            if pos is None:
-                m = (o != o0)
-                idx = m.nonzero()
-                pos = idx[:20]
-        if not torch.equal(lse, lse0):
-            print("pl", pl, "iter", i, "LSE differs too")
-    print(f"permlane={pl}: {diffs}/20 runs differ")
-    if pos is not None:
-        print("first mismatch positions (b,h,q,d):")
-        for r in pos.tolist():
-            print("  ", r, "q%64 =", r[2] % 64, "q%16 =", r[2] % 16,
-                  "d%16 =", r[3] % 16)
+o0, lse0 = ext.flash_attn_fwd_v3(q, k, v, 0.125, True)
+oc, lsec = o0.clone(), lse0.clone()
+b0 = ext.flash_attn_bwd_v3(dy, q, k, v, o0, lse0, 0.125, True)
+b0 = [t.clone() for t in b0]
 
-# v4 determinism too
-o0, _ = ext.flash_attn_fwd_v4(q, k, v, 0.125)
-d4 = sum(0 if torch.equal(ext.flash_attn_fwd_v4(q, k, v, 0.125)[0], o0)
-         else 1 for _ in range(20))
-print(f"v4: {d4}/20 runs differ")
-# v2 baseline
-o0, _ = ext.flash_attn_fwd(q, k, v, 0.125)
-d2 = sum(0 if torch.equal(ext.flash_attn_fwd(q, k, v, 0.125)[0], o0)
-         else 1 for _ in range(20))
-print(f"v2: {d2}/20 runs differ")
+for i in range(20):
+    o, lse = ext.flash_attn_fwd_v3(q, k, v, 0.125, True)
+    bb = ext.flash_attn_bwd_v3(dy, q, k, v, o, lse, 0.125, True)
+    for name, t, ref in (("q", q, qc), ("k", k, kc), ("v", v, vc),
+                         ("dy", dy, dyc), ("o", o, oc),
+                         ("lse", lse, lsec)):
+        if not torch.equal(t, ref):
+            n = (t != ref).sum().item()
+            idx = (t != ref).nonzero()[:6].tolist()
+            print(f"iter {i}: {name} changed at {n} positions, "
+                  f"e.g. {idx}")
+            break
+    for name, t, ref in zip(("dq", "dk", "dv"), bb, b0):
+        if not torch.equal(t, ref):
+            n = (t != ref).sum().item()
+            idx = (t != ref).nonzero()[:6].tolist()
+            print(f"iter {i}: {name} nondet at {n} positions, "
+                  f"e.g. {idx}")
+            break
+print("done")
