@@ -211,7 +211,7 @@ __device__ __forceinline__ bf16x8 read_frag_tr(const __bf16* img,
       "ds_read_b64_tr_b16 %0, %2\n"
       "ds_read_b64_tr_b16 %1, %3\n"
       "s_waitcnt lgkmcnt(0)"
-      : "=v"(lo), "=v"(hi)
+      : "=&v"(lo), "=&v"(hi)
       : "v"(a0), "v"(a1)
       : "memory");
   bf16x8 v;
@@ -248,7 +248,7 @@ __device__ __forceinline__ void read_frag_tr2(const __bf16* imgA,
       "ds_read_b64_tr_b16 %2, %6\n"
       "ds_read_b64_tr_b16 %3, %7\n"
       "s_waitcnt lgkmcnt(0)"
-      : "=v"(a0), "=v"(a1), "=v"(b0), "=v"(b1)
+      : "=&v"(a0), "=&v"(a1), "=&v"(b0), "=&v"(b1)
       : "v"(baseA + o0), "v"(baseA + o1), "v"(baseB + o0),
         "v"(baseB + o1)
       : "memory");
@@ -1527,13 +1527,47 @@ __device__ __forceinline__ bf16x8 read_frag_tr4(const __bf16* img,
       "ds_read_b64_tr_b16 %0, %2\n"
       "ds_read_b64_tr_b16 %1, %3\n"
       "s_waitcnt lgkmcnt(0)"
-      : "=v"(lo), "=v"(hi)
+      : "=&v"(lo), "=&v"(hi)
       : "v"(a0), "v"(a1)
       : "memory");
   bf16x8 v;
   unsigned* w = reinterpret_cast<unsigned*>(&v);
   w[0] = lo[0]; w[1] = lo[1]; w[2] = hi[0]; w[3] = hi[1];
   return v;
+}
+
+// Paired tr4: two adjacent kv chunks of the same 16-col block with a
+// single lgkm drain (8 serial drains per tile were parking v4 75%).
+__device__ __forceinline__ void read_frag_tr4x2(const __bf16* img,
+                                                int krow0a, int krow0b,
+                                                int col16, int lane,
+                                                bf16x8& va, bf16x8& vb) {
+  const int j = lane & 15;
+  const unsigned base =
+      (unsigned)(size_t)(__attribute__((address_space(3))) const char*)
+          (const void*)img;
+  const int cb = (col16 + 4 * (j & 3)) * 2;
+  const unsigned a0 = base + (unsigned)swz_off(krow0a + (j >> 2), cb);
+  const unsigned a1 =
+      base + (unsigned)swz_off(krow0a + 4 + (j >> 2), cb);
+  const unsigned b0 = base + (unsigned)swz_off(krow0b + (j >> 2), cb);
+  const unsigned b1 =
+      base + (unsigned)swz_off(krow0b + 4 + (j >> 2), cb);
+  typedef __attribute__((ext_vector_type(2))) unsigned uint2v;
+  uint2v ra0, ra1, rb0, rb1;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %4\n"
+      "ds_read_b64_tr_b16 %1, %5\n"
+      "ds_read_b64_tr_b16 %2, %6\n"
+      "ds_read_b64_tr_b16 %3, %7\n"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(ra0), "=&v"(ra1), "=&v"(rb0), "=&v"(rb1)
+      : "v"(a0), "v"(a1), "v"(b0), "v"(b1)
+      : "memory");
+  unsigned* wa = reinterpret_cast<unsigned*>(&va);
+  wa[0] = ra0[0]; wa[1] = ra0[1]; wa[2] = ra1[0]; wa[3] = ra1[1];
+  unsigned* wb = reinterpret_cast<unsigned*>(&vb);
+  wb[0] = rb0[0]; wb[1] = rb0[1]; wb[2] = rb1[0]; wb[3] = rb1[1];
 }
 
 __global__ __launch_bounds__(256, 4) void flash_fwd_v4_kernel(
@@ -1674,12 +1708,15 @@ __global__ __launch_bounds__(256, 4) void flash_fwd_v4_kernel(
       #pragma unroll
       for (int dt = 0; dt < 2; ++dt) {
         #pragma unroll
-        for (int t = 0; t < 4; ++t) {
-          bf16x8 va = read_frag_tr4(
-              lds_v[buf], 16 * t + 8 * (g >> 1),
-              32 * dt + 16 * (g & 1), lane);
+        for (int t = 0; t < 4; t += 2) {
+          bf16x8 va, vb;
+          read_frag_tr4x2(lds_v[buf], 16 * t + 8 * (g >> 1),
+                          16 * (t + 1) + 8 * (g >> 1),
+                          32 * dt + 16 * (g & 1), lane, va, vb);
           o_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               va, pb[t], o_acc[dt], 0, 0, 0);
+          o_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              vb, pb[t + 1], o_acc[dt], 0, 0, 0);
         }
       }
     }

@@ -303,7 +303,7 @@ __global__ void tr16_probe_kernel(unsigned short* __restrict__ out,
   asm volatile(
       "ds_read_b64_tr_b16 %0, %1\n"
       "s_waitcnt lgkmcnt(0)"
-      : "=v"(r)
+      : "=&v"(r)
       : "v"(base + (unsigned)addr)
       : "memory");
   out[lane * 4 + 0] = (unsigned short)(r[0] & 0xffff);
