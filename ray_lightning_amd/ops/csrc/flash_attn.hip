@@ -1113,8 +1113,7 @@ namespace {
 //   dV^T[d][kv]= mfma(dO^T A-frag(lds), redist(P) B-frag)
 //   dK^T[d][kv]= mfma(Q^T  A-frag(lds), redist(dS) B-frag)
 template <bool USE_PERMLANE>
-// min 4 waves/EU: LDS fits 4 workgroups, keep regs <=128
-__global__ __launch_bounds__(256, 4) void flash_bwd_dkv_v3_kernel(
+__global__ __launch_bounds__(256) void flash_bwd_dkv_v3_kernel(
     const __hip_bfloat16* __restrict__ Qg,
     const __hip_bfloat16* __restrict__ Kg,
     const __hip_bfloat16* __restrict__ Vg,
@@ -1143,8 +1142,7 @@ __global__ __launch_bounds__(256, 4) void flash_bwd_dkv_v3_kernel(
   __shared__ __bf16 lds_q[2][BN * HS];   // Q  row-major db, 16 KB
   __shared__ float lds_lse[2][BN];
   __shared__ float lds_d[2][BN];
-  // epilogue transposes reuse lds_do[0] (dead after the q loop):
-  // keeps LDS at 33.8 KB -> 4 workgroups/CU
+  __shared__ __bf16 lds_ep[4][WQ * HS];  // epilogue, 8 KB
 
   // K^T / V^T B-fragments (col = kv = lane&15): kernel-resident
   bf16x8 kf[2], vf[2];
@@ -1269,8 +1267,7 @@ __global__ __launch_bounds__(256, 4) void flash_bwd_dkv_v3_kernel(
   }
 
   // epilogue: transpose dV^T then dK^T through per-wave LDS
-  // (carved from lds_do[0]; the last loop barrier already fenced it)
-  __bf16* ew = lds_do[0] + wave * (WQ * HS);
+  __bf16* ew = lds_ep[wave];
   const int erow = lane >> 2;            // wave-local kv row
   const int ec0 = (lane & 3) * 16;
   #pragma unroll
@@ -1570,7 +1567,11 @@ __device__ __forceinline__ void read_frag_tr4x2(const __bf16* img,
   wb[0] = rb0[0]; wb[1] = rb0[1]; wb[2] = rb1[0]; wb[3] = rb1[1];
 }
 
-__global__ __launch_bounds__(256, 4) void flash_fwd_v4_kernel(
+// NOTE: no forced min-waves — occupancy pressure made the allocator
+// spill, and spill-reload of address registers around in-flight ops
+// produced stray writes (the r02 dk nondeterminism, profiles/
+// r02_flash_v3.md postmortem). Keep hand-asm kernels spill-free.
+__global__ __launch_bounds__(256) void flash_fwd_v4_kernel(
     const __hip_bfloat16* __restrict__ Q,
     const __hip_bfloat16* __restrict__ K,
     const __hip_bfloat16* __restrict__ V,
