@@ -10,7 +10,9 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from ray_lightning_amd import ops  # noqa: E402
 
 ext = ops._load_ext()
-B, H, T, hs = 8, 25, 1024, 64
+import os as _os
+B = int(_os.environ.get("FB_B", "8"))
+H, T, hs = 25, 1024, 64
 scale = 1.0 / math.sqrt(hs)
 torch.manual_seed(0)
 q = torch.randn(B, H, T, hs, device="cuda", dtype=torch.bfloat16)
