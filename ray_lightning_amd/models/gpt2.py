@@ -75,7 +75,11 @@ class MLP(nn.Module):
         self.c_proj = nn.Linear(4 * cfg.n_embd, cfg.n_embd)
 
     def forward(self, x):
-        return self.c_proj(F.gelu(self.c_fc(x), approximate="tanh"))
+        # hipBLASLt-epilogue fused path (GELU + bias grads inside the
+        # GEMMs) when usable; composed ops otherwise (ops/lt_mlp.py)
+        from ..ops.lt_mlp import fused_mlp
+        return fused_mlp(x, self.c_fc.weight, self.c_fc.bias,
+                         self.c_proj.weight, self.c_proj.bias)
 
 
 class Block(nn.Module):

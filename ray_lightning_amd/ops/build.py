@@ -36,6 +36,8 @@ def build(verbose: bool = False) -> None:
                       os.path.join(_CSRC, "mfma_probe.hip")], []),
         ("_rccl_comm", [os.path.join(_CSRC, "rccl_comm.hip")],
          ["-L/opt/rocm/lib", "-lrccl"]),
+        ("_lt_mlp", [os.path.join(_CSRC, "lt_mlp.hip")],
+         ["-L/opt/rocm/lib", "-lhipblaslt"]),
     ):
         build_dir = os.path.join(_BUILD, name)
         os.makedirs(build_dir, exist_ok=True)
