@@ -1,14 +1,18 @@
 // Fused transformer-MLP GEMMs via hipBLASLt epilogues (library GEMMs —
-// the "plain GEMM" tier; the point-wise GELU/dGELU/bias-grad work that
-// PyTorch runs as 4 separate memory-bound kernels per MLP is folded
-// into the GEMM epilogues instead):
-//   fwd:  h = GELU(x @ W1^T + b1)   [GELU_AUX_BIAS: stores pre-act z]
-//         y = h @ W2^T + b2         [BIAS]
-//   bwd:  dz = dGELU(dy @ W2, z), db1 = colsum(dz)  [DGELU_BGRAD]
-//         dW2 = dy^T @ h, db2 = rowsum(dy)          [BGRADB]
-//         dW1 = dz^T @ x; dx = dz @ W1              [plain]
-// Replaces: at::native GELU fwd+bwd kernels and two bias-grad
-// reduce_kernels per MLP per step (GPT-2-XL profile: ~5% of step).
+// the "plain GEMM" tier; the point-wise dGELU/bias-grad work that
+// PyTorch runs as separate memory-bound kernels per MLP is folded into
+// the GEMM epilogues where this hipBLASLt (1.2/gfx950) has algos —
+// probed with scripts/probe_lt.cpp: GELU_AUX_BIAS and DGELU_BGRAD have
+// NO algos; DGELU (NN) and BGRADB (NT wgrad) DO:
+//   fwd:  z = x @ W1^T + b1        [BIAS]
+//         h = gelu_tanh(z)         [one vectorized kernel here]
+//         y = h @ W2^T + b2        [BIAS]
+//   bwd:  dz = dGELU(dy @ W2, aux=z)            [DGELU]
+//         dW2 = dy^T @ h, db2 = colsum(dy)      [BGRADB]
+//         dW1 = dz^T @ x, db1 = colsum(dz)      [BGRADB]
+//         dx = dz @ W1                          [plain]
+// Replaces the GELU-backward kernel and both bias-grad reduce_kernels
+// per MLP per step (GPT-2-XL profile: ~3% of step).
 //
 // All tensors are row-major torch bf16; hipBLASLt is column-major, so
 // every call computes the transposed-view GEMM (the usual swap).
@@ -148,9 +152,28 @@ void lt_gemm(int64_t m, int64_t n, int64_t k, hipblasOperation_t opA,
 inline const void* dp(const torch::Tensor& t) { return t.data_ptr(); }
 inline void* dp(torch::Tensor& t) { return t.data_ptr(); }
 
+using bf16x8v = __attribute__((ext_vector_type(8))) __bf16;
+
+// h = gelu_tanh(z), 8-wide bf16
+__global__ __launch_bounds__(256) void gelu_fwd_kernel(
+    const __hip_bfloat16* __restrict__ z, __hip_bfloat16* __restrict__ h,
+    long n8) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n8) return;
+  bf16x8v zv = *reinterpret_cast<const bf16x8v*>(z + i * 8);
+  bf16x8v hv;
+  #pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const float x = (float)zv[j];
+    const float c = 0.7978845608028654f * (x + 0.044715f * x * x * x);
+    hv[j] = (__bf16)(0.5f * x * (1.0f + tanhf(c)));
+  }
+  *reinterpret_cast<bf16x8v*>(h + i * 8) = hv;
+}
+
 }  // namespace
 
-// fwd: returns {h, y, z_aux}
+// fwd: returns {y, h, z}
 std::vector<torch::Tensor> lt_mlp_fwd(torch::Tensor x,
                                       torch::Tensor W1,
                                       torch::Tensor b1,
@@ -160,14 +183,21 @@ std::vector<torch::Tensor> lt_mlp_fwd(torch::Tensor x,
               x.dim() == 2 && x.is_contiguous());
   const int64_t M = x.size(0), C = x.size(1), F = W1.size(0);
   TORCH_CHECK(W1.size(1) == C && W2.size(0) == C && W2.size(1) == F);
+  TORCH_CHECK((M * F) % 8 == 0);
   auto h = torch::empty({M, F}, x.options());
   auto z = torch::empty({M, F}, x.options());
   auto y = torch::empty({M, C}, x.options());
   auto stream = at::hip::getCurrentHIPStream().stream();
-  // h_cm[F,M] = W1_cm[C,F]^T @ x_cm[C,M]; aux z same layout as D
+  // z_cm[F,M] = W1_cm[C,F]^T @ x_cm[C,M] + b1
   lt_gemm(F, M, C, HIPBLAS_OP_T, HIPBLAS_OP_N, dp(W1), C, dp(x), C,
-          dp(h), F, HIPBLASLT_EPILOGUE_GELU_AUX_BIAS, dp(b1), dp(z), F,
+          dp(z), F, HIPBLASLT_EPILOGUE_BIAS, dp(b1), nullptr, 0,
           stream);
+  const long n8 = (long)M * F / 8;
+  hipLaunchKernelGGL(gelu_fwd_kernel, dim3((n8 + 255) / 256), dim3(256),
+                     0, stream,
+                     reinterpret_cast<const __hip_bfloat16*>(z.data_ptr()),
+                     reinterpret_cast<__hip_bfloat16*>(h.data_ptr()),
+                     n8);
   // y_cm[C,M] = W2_cm[F,C]^T @ h_cm[F,M]
   lt_gemm(C, M, F, HIPBLAS_OP_T, HIPBLAS_OP_N, dp(W2), F, dp(h), F,
           dp(y), C, HIPBLASLT_EPILOGUE_BIAS, dp(b2), nullptr, 0,
@@ -189,17 +219,17 @@ std::vector<torch::Tensor> lt_mlp_bwd(torch::Tensor dy, torch::Tensor x,
   auto db1 = torch::empty({F}, x.options());
   auto db2 = torch::empty({C}, x.options());
   auto stream = at::hip::getCurrentHIPStream().stream();
-  // dz_cm[F,M] = W2_cm[F,C] @ dy_cm[C,M], dGELU(aux=z), db1 = bgrad
+  // dz_cm[F,M] = W2_cm[F,C] @ dy_cm[C,M], then dGELU with aux=z
   lt_gemm(F, M, C, HIPBLAS_OP_N, HIPBLAS_OP_N, dp(W2), F, dp(dy), C,
-          dp(dz), F, HIPBLASLT_EPILOGUE_DGELU_BGRAD, dp(db1), dp(z), F,
+          dp(dz), F, HIPBLASLT_EPILOGUE_DGELU, nullptr, dp(z), F,
           stream);
-  // dW2_rm[C,F] -> cm[F,C] = h_cm[F,M] @ dy_cm[C,M]^T; db2 = rowsum(dy)
+  // dW2_rm[C,F] -> cm[F,C] = h_cm[F,M] @ dy_cm[C,M]^T; db2 from B=dy
   lt_gemm(F, C, M, HIPBLAS_OP_N, HIPBLAS_OP_T, dp(h), F, dp(dy), C,
           dp(dW2), F, HIPBLASLT_EPILOGUE_BGRADB, dp(db2), nullptr, 0,
           stream);
-  // dW1_rm[F,C] -> cm[C,F] = x_cm[C,M] @ dz_cm[F,M]^T
+  // dW1_rm[F,C] -> cm[C,F] = x_cm[C,M] @ dz_cm[F,M]^T; db1 from B=dz
   lt_gemm(C, F, M, HIPBLAS_OP_N, HIPBLAS_OP_T, dp(x), C, dp(dz), F,
-          dp(dW1), C, HIPBLASLT_EPILOGUE_DEFAULT, nullptr, nullptr, 0,
+          dp(dW1), C, HIPBLASLT_EPILOGUE_BGRADB, dp(db1), nullptr, 0,
           stream);
   // dx_cm[C,M] = W1_cm[C,F] @ dz_cm[F,M]
   lt_gemm(C, M, F, HIPBLAS_OP_N, HIPBLAS_OP_N, dp(W1), C, dp(dz), F,
