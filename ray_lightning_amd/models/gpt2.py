@@ -54,7 +54,8 @@ class CausalSelfAttention(nn.Module):
 
     def forward(self, x):
         B, T, C = x.shape
-        qkv = self.c_attn(x)
+        from ..ops.lt_mlp import lt_linear
+        qkv = lt_linear(x, self.c_attn.weight, self.c_attn.bias)
         q, k, v = qkv.split(C, dim=2)
         hs = C // self.n_head
         q = q.view(B, T, self.n_head, hs).transpose(1, 2)
@@ -65,7 +66,7 @@ class CausalSelfAttention(nn.Module):
         from ..ops.flash_attn import flash_attention
         y = flash_attention(q, k, v)
         y = y.transpose(1, 2).contiguous().view(B, T, C)
-        return self.c_proj(y)
+        return lt_linear(y, self.c_proj.weight, self.c_proj.bias)
 
 
 class MLP(nn.Module):

@@ -238,9 +238,50 @@ std::vector<torch::Tensor> lt_mlp_bwd(torch::Tensor dy, torch::Tensor x,
   return {dx, dW1, db1, dW2, db2};
 }
 
+
+// General Linear: y = x @ W^T + b with BIAS fwd epilogue; backward
+// fuses db = colsum(dy) into the wgrad GEMM via BGRADB.
+std::vector<torch::Tensor> lt_linear_fwd(torch::Tensor x,
+                                         torch::Tensor W,
+                                         torch::Tensor b) {
+  TORCH_CHECK(x.is_cuda() && x.dtype() == at::kBFloat16 &&
+              x.dim() == 2 && x.is_contiguous());
+  const int64_t M = x.size(0), C = x.size(1), N = W.size(0);
+  TORCH_CHECK(W.size(1) == C);
+  auto y = torch::empty({M, N}, x.options());
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  lt_gemm(N, M, C, HIPBLAS_OP_T, HIPBLAS_OP_N, dp(W), C, dp(x), C,
+          dp(y), N, HIPBLASLT_EPILOGUE_BIAS, dp(b), nullptr, 0,
+          stream);
+  return {y};
+}
+
+std::vector<torch::Tensor> lt_linear_bwd(torch::Tensor dy,
+                                         torch::Tensor x,
+                                         torch::Tensor W) {
+  const int64_t M = x.size(0), C = x.size(1), N = W.size(0);
+  dy = dy.contiguous();
+  auto dx = torch::empty({M, C}, x.options());
+  auto dW = torch::empty({N, C}, x.options());
+  auto db = torch::empty({N}, x.options());
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  // dW_rm[N,C] -> cm[C,N] = x_cm[C,M] @ dy_cm[N,M]^T; db from B=dy
+  lt_gemm(C, N, M, HIPBLAS_OP_N, HIPBLAS_OP_T, dp(x), C, dp(dy), N,
+          dp(dW), C, HIPBLASLT_EPILOGUE_BGRADB, dp(db), nullptr, 0,
+          stream);
+  // dx_cm[C,M] = W_cm[C,N] @ dy_cm[N,M]
+  lt_gemm(C, M, N, HIPBLAS_OP_N, HIPBLAS_OP_N, dp(W), C, dp(dy), N,
+          dp(dx), C, HIPBLASLT_EPILOGUE_DEFAULT, nullptr, nullptr, 0,
+          stream);
+  return {dx, dW, db};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("lt_mlp_fwd", &lt_mlp_fwd,
         "fused MLP forward (GELU_AUX_BIAS + BIAS epilogues)");
+  m.def("lt_linear_fwd", &lt_linear_fwd, "Linear fwd (BIAS epilogue)");
+  m.def("lt_linear_bwd", &lt_linear_bwd,
+        "Linear bwd (BGRADB-fused bias grad)");
   m.def("lt_mlp_bwd", &lt_mlp_bwd,
         "fused MLP backward (DGELU_BGRAD + BGRADB epilogues)");
 }

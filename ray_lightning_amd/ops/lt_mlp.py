@@ -64,3 +64,29 @@ def fused_mlp(x: torch.Tensor, W1: torch.Tensor, b1: torch.Tensor,
         return out.reshape(*shp[:-1], -1)
     return F.linear(F.gelu(F.linear(x, W1, b1), approximate="tanh"),
                     W2, b2)
+
+
+class _LtLinear(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, W, b):
+        (y,) = _load().lt_linear_fwd(x, W, b)
+        ctx.save_for_backward(x, W)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, W = ctx.saved_tensors
+        dx, dW, db = _load().lt_linear_bwd(dy, x, W)
+        return dx, dW, db
+
+
+def lt_linear(x: torch.Tensor, W: torch.Tensor,
+              b: torch.Tensor) -> torch.Tensor:
+    """nn.Linear-equivalent with the bias gradient fused into the wgrad
+    GEMM (BGRADB). Falls back to F.linear off the fast path."""
+    if (lt_mlp_available() and x.is_cuda and b is not None
+            and x.dtype == torch.bfloat16 and W.dtype == torch.bfloat16):
+        shp = x.shape
+        out = _LtLinear.apply(x.reshape(-1, shp[-1]).contiguous(), W, b)
+        return out.reshape(*shp[:-1], -1)
+    return F.linear(x, W, b)

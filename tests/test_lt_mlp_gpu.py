@@ -79,3 +79,24 @@ def test_lt_mlp_in_gpt2_block():
         scale = gs[n].abs().max().item() + 1e-6
         err = (gf[n] - gs[n]).abs().max().item()
         assert err < 7e-2 * max(scale, 1.0), f"{n}: {err}"
+
+
+def test_lt_linear_matches_reference():
+    from ray_lightning_amd.ops.lt_mlp import lt_linear
+    torch.manual_seed(3)
+    M, C, N = 512, 256, 768
+    x = (torch.randn(M, C, device="cuda") * 0.5).bfloat16()
+    W = (torch.randn(N, C, device="cuda") * 0.05).bfloat16()
+    b = torch.randn(N, device="cuda").bfloat16() * 0.1
+    xs = [t.clone().requires_grad_(True) for t in (x, W, b)]
+    y = lt_linear(*xs)
+    dy = torch.randn_like(y) * 0.5
+    y.backward(dy)
+    rs = [t.clone().float().requires_grad_(True) for t in (x, W, b)]
+    ref = torch.nn.functional.linear(*rs)
+    ref.backward(dy.float())
+    assert torch.allclose(y.float(), ref, atol=5e-2, rtol=3e-2)
+    for name, got, want in zip(("dx", "dW", "db"), xs, rs):
+        scale = want.grad.abs().max().item() + 1e-6
+        err = (got.grad.float() - want.grad).abs().max().item()
+        assert err < 4e-2 * max(scale, 1.0), f"{name}: {err}"
