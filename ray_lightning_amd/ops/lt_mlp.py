@@ -1,10 +1,15 @@
-"""Fused transformer MLP on hipBLASLt epilogues.
+"""Fused transformer MLP on hipBLASLt epilogues — EXPERIMENTAL, off by
+default (``RLA_LT_MLP=1`` opts in).
 
 ``fused_mlp(x, W1, b1, W2, b2)`` == ``Linear2(gelu_tanh(Linear1(x)))``
-with the GELU forward/backward and both bias-gradient reductions folded
-into the GEMM epilogues (csrc/lt_mlp.hip) — removes four memory-bound
-point-wise/reduce kernels per MLP per step. Falls back to the composed
-torch ops off-GPU / non-bf16. ``RLA_LT_MLP=0`` disables.
+with dGELU and both bias-gradient reductions folded into the GEMM
+epilogues (csrc/lt_mlp.hip). Measured verdict on hipBLASLt 1.2/gfx950
+(profiles/r02_lt_epilogues.md): this library build's epilogue-fused
+NT-wgrad kernels are all tiny-macro-tile (MT32x32) variants — 2.5 ms
+vs 0.6 ms per GPT-2-XL wgrad — so the fusion костs far more GEMM time
+than the ~7 ms/step of point-wise kernels it removes (59.9k -> 26.7k
+tokens/s end to end). Kept as a correct, tested path so the fusion can
+be re-evaluated on future hipBLASLt releases.
 """
 from __future__ import annotations
 
@@ -34,7 +39,7 @@ def _load():
 
 def lt_mlp_available() -> bool:
     return (torch.cuda.is_available() and _load() is not None
-            and os.environ.get("RLA_LT_MLP", "1") != "0")
+            and os.environ.get("RLA_LT_MLP", "0") == "1")
 
 
 class _FusedMLP(torch.autograd.Function):

@@ -7,6 +7,10 @@ pytestmark = pytest.mark.gpu
 if not torch.cuda.is_available():
     pytest.skip("needs MI355X", allow_module_level=True)
 
+import os
+
+os.environ["RLA_LT_MLP"] = "1"  # experimental path under test
+
 from ray_lightning_amd.ops.lt_mlp import _load, fused_mlp
 
 assert _load() is not None, "_lt_mlp extension failed to load"
