@@ -675,6 +675,103 @@ std::vector<torch::Tensor> fused_ln_bwd(
     c10::optional<torch::Tensor> dext, torch::Tensor gamma,
     torch::Tensor mean, torch::Tensor invstd);
 
+
+// ===================================================================
+// fused dGELU(tanh) + bias-gradient partials: one pass over [M, F]
+// computes dz = dh * gelu'(z) AND per-slice column sums of dz, so the
+// separate GeluBackward kernel + the full reduce_kernel re-read of dz
+// collapse into one streaming pass + a tiny fold (the deterministic
+// two-stage pattern from the fused-BN kernels).
+// ===================================================================
+namespace {
+
+__global__ __launch_bounds__(256) void dgelu_bgrad_kernel(
+    const __hip_bfloat16* __restrict__ dh,
+    const __hip_bfloat16* __restrict__ z,
+    __hip_bfloat16* __restrict__ dz, float* __restrict__ partial,
+    long M, long F, long rows_per_slice) {
+  // block: 256 threads x 2 cols each = 512 columns; grid.x = col
+  // blocks, grid.y = row slices
+  const long c0 = (long)blockIdx.x * 512 + threadIdx.x * 2;
+  if (c0 >= F) return;
+  const long r0 = (long)blockIdx.y * rows_per_slice;
+  const long r1 = (r0 + rows_per_slice < M) ? r0 + rows_per_slice : M;
+  const bool two = (c0 + 1) < F;
+  float s0 = 0.f, s1 = 0.f;
+  for (long r = r0; r < r1; ++r) {
+    const long off = r * F + c0;
+    const float zv0 = (float)z[off];
+    const float g0 = (float)dh[off];
+    // gelu'(x) for tanh approximation
+    const float t0 = tanhf(0.7978845608028654f *
+                           (zv0 + 0.044715f * zv0 * zv0 * zv0));
+    const float d0 =
+        0.5f * (1.0f + t0) +
+        0.5f * zv0 * (1.0f - t0 * t0) * 0.7978845608028654f *
+            (1.0f + 0.134145f * zv0 * zv0);
+    const float o0 = g0 * d0;
+    dz[off] = (__hip_bfloat16)o0;
+    s0 += o0;
+    if (two) {
+      const float zv1 = (float)z[off + 1];
+      const float g1 = (float)dh[off + 1];
+      const float t1 = tanhf(0.7978845608028654f *
+                             (zv1 + 0.044715f * zv1 * zv1 * zv1));
+      const float d1 =
+          0.5f * (1.0f + t1) +
+          0.5f * zv1 * (1.0f - t1 * t1) * 0.7978845608028654f *
+              (1.0f + 0.134145f * zv1 * zv1);
+      const float o1 = g1 * d1;
+      dz[off + 1] = (__hip_bfloat16)o1;
+      s1 += o1;
+    }
+  }
+  float* prow = partial + (long)blockIdx.y * F;
+  prow[c0] = s0;
+  if (two) prow[c0 + 1] = s1;
+}
+
+__global__ __launch_bounds__(256) void bgrad_fold_kernel(
+    const float* __restrict__ partial, __hip_bfloat16* __restrict__ db,
+    long F, long slices) {
+  const long c = (long)blockIdx.x * 256 + threadIdx.x;
+  if (c >= F) return;
+  float s = 0.f;
+  for (long i = 0; i < slices; ++i) s += partial[i * F + c];
+  db[c] = (__hip_bfloat16)s;
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> fused_dgelu_bgrad(torch::Tensor dh,
+                                             torch::Tensor z) {
+  TORCH_CHECK(dh.is_cuda() && dh.scalar_type() == at::kBFloat16 &&
+              dh.is_contiguous() && z.is_contiguous() &&
+              dh.sizes() == z.sizes() && dh.dim() == 2);
+  const long M = dh.size(0), F = dh.size(1);
+  auto dz = torch::empty_like(dh);
+  auto db = torch::empty({F}, dh.options());
+  const long rows_per_slice = 512;
+  const long slices = (M + rows_per_slice - 1) / rows_per_slice;
+  auto partial = torch::empty({slices, F},
+                              dh.options().dtype(at::kFloat));
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  dim3 grid((F + 511) / 512, slices);
+  hipLaunchKernelGGL(dgelu_bgrad_kernel, grid, dim3(256), 0, stream,
+      reinterpret_cast<const __hip_bfloat16*>(dh.data_ptr()),
+      reinterpret_cast<const __hip_bfloat16*>(z.data_ptr()),
+      reinterpret_cast<__hip_bfloat16*>(dz.data_ptr()),
+      partial.data_ptr<float>(), M, F, rows_per_slice);
+  hipLaunchKernelGGL(bgrad_fold_kernel, dim3((F + 255) / 256),
+                     dim3(256), 0, stream, partial.data_ptr<float>(),
+                     reinterpret_cast<__hip_bfloat16*>(db.data_ptr()),
+                     F, slices);
+  hipError_t e = hipGetLastError();
+  TORCH_CHECK(e == hipSuccess, "fused_dgelu_bgrad: ",
+              hipGetErrorString(e));
+  return {dz, db};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("multi_tensor_pack", &multi_tensor_pack,
         "multi-tensor flatten/cast into bucket slots");
@@ -706,6 +803,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "ds_read_b64_tr_b16 lane/element map probe");
   m.def("permlane_swap_probe", &permlane_swap_probe,
         "permlane{16,32}_swap lane-exchange semantics probe");
+  m.def("fused_dgelu_bgrad", &fused_dgelu_bgrad,
+        "dz = dh * gelu_tanh'(z) + column-sum bias grad, one pass");
   m.def("scale_inplace", &scale_inplace, "flat *= s");
   m.def("scale_cast", &scale_cast, "dst_f32 = src_bf16 * s");
   m.def("fused_sgd", &fused_sgd, "fused multi-tensor SGD(momentum)");

@@ -1,0 +1,55 @@
+"""GPT-2 MLP with the hand-written fused dGELU + bias-grad kernel.
+
+Forward is the standard tuned-GEMM path (torch addmm -> hipBLASLt);
+the backward replaces at::native's GeluBackward kernel plus the full
+reduce_kernel re-read of dz with ONE streaming pass
+(csrc/hip_ops.hip::fused_dgelu_bgrad) and computes the remaining grads
+with the same GEMMs autograd would issue. ``RLA_FUSED_MLP=0`` disables.
+"""
+from __future__ import annotations
+
+import os
+
+import torch
+import torch.nn.functional as F
+
+from . import _load_ext
+
+
+def _usable(x: torch.Tensor, W1: torch.Tensor) -> bool:
+    return (os.environ.get("RLA_FUSED_MLP", "1") != "0" and x.is_cuda
+            and x.dtype == torch.bfloat16
+            and W1.dtype == torch.bfloat16 and _load_ext() is not None)
+
+
+class _MLP(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, W1, b1, W2, b2):
+        z = torch.addmm(b1, x, W1.t())
+        h = F.gelu(z, approximate="tanh")
+        y = torch.addmm(b2, h, W2.t())
+        ctx.save_for_backward(x, W1, W2, z, h)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, W1, W2, z, h = ctx.saved_tensors
+        dy = dy.contiguous()
+        dh = dy @ W2
+        dz, db1 = _load_ext().fused_dgelu_bgrad(dh, z)
+        dW1 = dz.t() @ x
+        dx = dz @ W1
+        dW2 = dy.t() @ h
+        db2 = dy.sum(dim=0)
+        return dx, dW1, db1, dW2, db2
+
+
+def fused_mlp(x: torch.Tensor, W1: torch.Tensor, b1: torch.Tensor,
+              W2: torch.Tensor, b2: torch.Tensor) -> torch.Tensor:
+    if _usable(x, W1):
+        shp = x.shape
+        out = _MLP.apply(x.reshape(-1, shp[-1]).contiguous(), W1, b1,
+                         W2, b2)
+        return out.reshape(*shp[:-1], -1)
+    return F.linear(F.gelu(F.linear(x, W1, b1), approximate="tanh"),
+                    W2, b2)

@@ -104,3 +104,52 @@ def test_lt_linear_matches_reference():
         scale = want.grad.abs().max().item() + 1e-6
         err = (got.grad.float() - want.grad).abs().max().item()
         assert err < 4e-2 * max(scale, 1.0), f"{name}: {err}"
+
+
+def test_fused_dgelu_bgrad_numerics():
+    """Hand-written fused dGELU+bias-grad vs the composed reference."""
+    from ray_lightning_amd.ops import _load_ext
+    ext = _load_ext()
+    torch.manual_seed(5)
+    M, F = 1024, 800
+    dh = (torch.randn(M, F, device="cuda") * 0.5).bfloat16()
+    z = (torch.randn(M, F, device="cuda")).bfloat16()
+    dz, db = ext.fused_dgelu_bgrad(dh, z)
+    zf = z.float().requires_grad_(True)
+    hf = torch.nn.functional.gelu(zf, approximate="tanh")
+    hf.backward(dh.float())
+    ref_dz = zf.grad
+    assert torch.allclose(dz.float(), ref_dz, atol=2e-2, rtol=2e-2), \
+        f"dz max err {(dz.float() - ref_dz).abs().max()}"
+    ref_db = ref_dz.sum(dim=0)
+    scale = ref_db.abs().max().item() + 1e-6
+    err = (db.float() - ref_db).abs().max().item()
+    assert err < 3e-2 * scale, f"db max err {err} scale {scale}"
+
+
+def test_fused_mlp_autograd_matches_reference():
+    from ray_lightning_amd.ops.mlp import fused_mlp
+    torch.manual_seed(6)
+    M, C, F = 512, 256, 1024
+    x = (torch.randn(M, C, device="cuda") * 0.5).bfloat16()
+    W1 = (torch.randn(F, C, device="cuda") * 0.05).bfloat16()
+    b1 = torch.randn(F, device="cuda").bfloat16() * 0.1
+    W2 = (torch.randn(C, F, device="cuda") * 0.05).bfloat16()
+    b2 = torch.randn(C, device="cuda").bfloat16() * 0.1
+    xs = [t.clone().requires_grad_(True) for t in (x, W1, b1, W2, b2)]
+    y = fused_mlp(*xs)
+    dy = torch.randn_like(y) * 0.5
+    y.backward(dy)
+    rs = [t.clone().float().requires_grad_(True)
+          for t in (x, W1, b1, W2, b2)]
+    ref = torch.nn.functional.linear(
+        torch.nn.functional.gelu(
+            torch.nn.functional.linear(rs[0], rs[1], rs[2]),
+            approximate="tanh"), rs[3], rs[4])
+    ref.backward(dy.float())
+    assert torch.allclose(y.float(), ref, atol=8e-2, rtol=4e-2)
+    for name, got, want in zip(("dx", "dW1", "db1", "dW2", "db2"),
+                               xs, rs):
+        scale = want.grad.abs().max().item() + 1e-6
+        err = (got.grad.float() - want.grad).abs().max().item()
+        assert err < 5e-2 * max(scale, 1.0), f"{name}: {err}"
