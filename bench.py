@@ -90,6 +90,10 @@ def parse_args() -> argparse.Namespace:
     p.add_argument("--memory-format", choices=["channels_last", "nchw"],
                    default="channels_last",
                    help="NHWC is the native MIOpen/CDNA4 conv layout")
+    p.add_argument("--hip-graphs", action="store_true",
+                   help="capture the train step in hipGraphs (engine "
+                        "mode, world=1): one replay per step instead "
+                        "of ~800 kernel launches")
     p.add_argument("--nccl-debug", action="store_true",
                    help="print per-rank RCCL ring/channel topology "
                         "(NCCL_DEBUG=INFO) so multi-GPU runs are "
@@ -388,6 +392,31 @@ def run_engine_mode(args) -> None:
 
     for i in range(args.warmup):
         step(i)
+
+    if args.hip_graphs and on_gpu and world == 1:
+        # capture one graph per rotating buffer (keeps the exact data
+        # rotation) after the warmup has settled MIOpen/hipBLASLt algos
+        try:
+            torch.cuda.synchronize()
+            graphs = []
+            for i in range(n_buf):
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    step(i)
+                graphs.append(g)
+            torch.cuda.synchronize()
+            eager_step = step
+
+            def step(i):  # noqa: F811 — replay replaces the eager step
+                graphs[i % n_buf].replay()
+
+            if rank == 0:
+                print("[bench] hipGraph capture OK "
+                      f"({n_buf} graphs)", flush=True)
+        except Exception as e:  # noqa: BLE001 — fall back to eager
+            if rank == 0:
+                print(f"[bench] hipGraph capture failed ({e}); "
+                      "eager steps", flush=True)
 
     if control is not None:
         control.barrier()
