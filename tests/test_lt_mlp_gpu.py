@@ -9,7 +9,8 @@ if not torch.cuda.is_available():
 
 import os
 
-os.environ["RLA_LT_MLP"] = "1"  # experimental path under test
+os.environ["RLA_LT_MLP"] = "1"  # experimental paths under test
+os.environ["RLA_FUSED_MLP"] = "1"
 
 from ray_lightning_amd.ops.lt_mlp import _load, fused_mlp
 
