@@ -1106,6 +1106,60 @@ std::vector<torch::Tensor> flash_attn_fwd_v3(torch::Tensor q,
 // =====================================================================
 namespace {
 
+
+// Batched variant for dkv: TWO 16-row blocks' transposed fragments
+// from two images with ONE lgkm drain (half the drains of per-n
+// reads; the full 4-block batch costs ~195 regs -> 2 waves/SIMD,
+// worse than the LDS-capped 3).
+__device__ __forceinline__ void read_frag_tr2x2(
+    const __bf16* imgA, const __bf16* imgB, int krow0, int n0,
+    int lane, bf16x8 (&va)[2], bf16x8 (&vb)[2]) {
+  const int j = lane & 15;
+  const int g = lane >> 4;
+  const unsigned baseA =
+      (unsigned)(size_t)(__attribute__((address_space(3))) const char*)
+          (const void*)imgA;
+  const unsigned baseB =
+      (unsigned)(size_t)(__attribute__((address_space(3))) const char*)
+          (const void*)imgB;
+  const int r0 = krow0 + 8 * g + (j >> 2);
+  typedef __attribute__((ext_vector_type(2))) unsigned uint2v;
+  uint2v a[4], b[4];
+  unsigned addrA[4], addrB[4];
+  #pragma unroll
+  for (int n = 0; n < 2; ++n) {
+    const int cb = (16 * (n0 + n) + 4 * (j & 3)) * 2;
+    addrA[2 * n] = baseA + (unsigned)swz_off(r0, cb);
+    addrA[2 * n + 1] = baseA + (unsigned)swz_off(r0 + 4, cb);
+    addrB[2 * n] = baseB + (unsigned)swz_off(r0, cb);
+    addrB[2 * n + 1] = baseB + (unsigned)swz_off(r0 + 4, cb);
+  }
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %8\n"
+      "ds_read_b64_tr_b16 %1, %9\n"
+      "ds_read_b64_tr_b16 %2, %10\n"
+      "ds_read_b64_tr_b16 %3, %11\n"
+      "ds_read_b64_tr_b16 %4, %12\n"
+      "ds_read_b64_tr_b16 %5, %13\n"
+      "ds_read_b64_tr_b16 %6, %14\n"
+      "ds_read_b64_tr_b16 %7, %15\n"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(a[0]), "=&v"(a[1]), "=&v"(a[2]), "=&v"(a[3]),
+        "=&v"(b[0]), "=&v"(b[1]), "=&v"(b[2]), "=&v"(b[3])
+      : "v"(addrA[0]), "v"(addrA[1]), "v"(addrA[2]), "v"(addrA[3]),
+        "v"(addrB[0]), "v"(addrB[1]), "v"(addrB[2]), "v"(addrB[3])
+      : "memory");
+  #pragma unroll
+  for (int n = 0; n < 2; ++n) {
+    unsigned* wa = reinterpret_cast<unsigned*>(&va[n]);
+    wa[0] = a[2 * n][0]; wa[1] = a[2 * n][1];
+    wa[2] = a[2 * n + 1][0]; wa[3] = a[2 * n + 1][1];
+    unsigned* wb = reinterpret_cast<unsigned*>(&vb[n]);
+    wb[0] = b[2 * n][0]; wb[1] = b[2 * n][1];
+    wb[2] = b[2 * n + 1][0]; wb[3] = b[2 * n + 1][1];
+  }
+}
+
 // dK/dV v3: workgroup = 64 KV rows, 4 waves x 16 kv columns; iterate
 // q tiles of 64. C layouts put kv in lane&15 throughout:
 //   S[q][kv]   = mfma(Q_frag,  K^T B-frag)   (both direct row-major)
@@ -1113,7 +1167,11 @@ namespace {
 //   dV^T[d][kv]= mfma(dO^T A-frag(lds), redist(P) B-frag)
 //   dK^T[d][kv]= mfma(Q^T  A-frag(lds), redist(dS) B-frag)
 template <bool USE_PERMLANE>
-__global__ __launch_bounds__(256) void flash_bwd_dkv_v3_kernel(
+// min 3 waves/EU (the LDS cap): the pair-batched tr reads push regs to
+// ~172 which rounds to 2 waves without the bound. Postmortem rule:
+// verified ScratchSize stays 0 under this bound (spills + hand asm
+// were the r02 nondeterminism bug).
+__global__ __launch_bounds__(256, 3) void flash_bwd_dkv_v3_kernel(
     const __hip_bfloat16* __restrict__ Qg,
     const __hip_bfloat16* __restrict__ Kg,
     const __hip_bfloat16* __restrict__ Vg,
@@ -1242,17 +1300,19 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_v3_kernel(
           unsigned* pw = reinterpret_cast<unsigned*>(&db);
           pw[0] = d0; pw[1] = d1; pw[2] = d2; pw[3] = d3;
         }
+        // d-blocks in pairs: dO^T / Q^T fragments, one drain per pair
         #pragma unroll
-        for (int n = 0; n < 4; ++n) {
-          // dO^T / Q^T fragments (row d, k = q) via hardware-transpose
-          // reads from the same row-major images (one lgkm drain)
-          bf16x8 adot, aqt;
-          read_frag_tr2(lds_do[buf], lds_q[buf], 32 * blk, 16 * n,
-                        lane, adot, aqt);
-          dv_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              adot, pb, dv_acc[n], 0, 0, 0);
-          dk_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              aqt, db, dk_acc[n], 0, 0, 0);
+        for (int np = 0; np < 2; ++np) {
+          bf16x8 adot[2], aqt[2];
+          read_frag_tr2x2(lds_do[buf], lds_q[buf], 32 * blk, 2 * np,
+                          lane, adot, aqt);
+          #pragma unroll
+          for (int n = 0; n < 2; ++n) {
+            dv_acc[2 * np + n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                adot[n], pb, dv_acc[2 * np + n], 0, 0, 0);
+            dk_acc[2 * np + n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                aqt[n], db, dk_acc[2 * np + n], 0, 0, 0);
+          }
         }
       }
     }
