@@ -220,6 +220,43 @@ __device__ __forceinline__ bf16x8 read_frag_tr(const __bf16* img,
   return v;
 }
 
+// Paired variant for the fwd PV loop: BOTH kv chunks' transposed
+// fragments of one 16-col block from one image, single lgkm drain
+// (separate absolute row bases keep the swizzle sel correct).
+__device__ __forceinline__ void read_frag_tr_2row(const __bf16* img,
+                                                  int krow0a,
+                                                  int krow0b, int col16,
+                                                  int lane, bf16x8& va,
+                                                  bf16x8& vb) {
+  const int j = lane & 15;
+  const int g = lane >> 4;
+  const unsigned base =
+      (unsigned)(size_t)(__attribute__((address_space(3))) const char*)
+          (const void*)img;
+  const int cb = (col16 + 4 * (j & 3)) * 2;
+  const int ra = krow0a + 8 * g + (j >> 2);
+  const int rb = krow0b + 8 * g + (j >> 2);
+  const unsigned a0 = base + (unsigned)swz_off(ra, cb);
+  const unsigned a1 = base + (unsigned)swz_off(ra + 4, cb);
+  const unsigned b0 = base + (unsigned)swz_off(rb, cb);
+  const unsigned b1 = base + (unsigned)swz_off(rb + 4, cb);
+  typedef __attribute__((ext_vector_type(2))) unsigned uint2v;
+  uint2v ra0, ra1, rb0, rb1;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %4\n"
+      "ds_read_b64_tr_b16 %1, %5\n"
+      "ds_read_b64_tr_b16 %2, %6\n"
+      "ds_read_b64_tr_b16 %3, %7\n"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(ra0), "=&v"(ra1), "=&v"(rb0), "=&v"(rb1)
+      : "v"(a0), "v"(a1), "v"(b0), "v"(b1)
+      : "memory");
+  unsigned* wa = reinterpret_cast<unsigned*>(&va);
+  wa[0] = ra0[0]; wa[1] = ra0[1]; wa[2] = ra1[0]; wa[3] = ra1[1];
+  unsigned* wb = reinterpret_cast<unsigned*>(&vb);
+  wb[0] = rb0[0]; wb[1] = rb0[1]; wb[2] = rb1[0]; wb[3] = rb1[1];
+}
+
 // Paired variant: transposed fragments from TWO images with a single
 // lgkm drain (dkv reads dO^T and Q^T back to back — separate helper
 // calls would serialize on their internal waits).
@@ -1025,16 +1062,16 @@ __global__ __launch_bounds__(256, 4) void flash_fwd_v3_kernel(
         unsigned* pw = reinterpret_cast<unsigned*>(&pb[blk]);
         pw[0] = d0; pw[1] = d1; pw[2] = d2; pw[3] = d3;
       }
-      // --- O^T += V^T P^T -------------------------------------------
+      // --- O^T += V^T P^T: both kv chunks' V^T fragments per
+      // d-block with one drain
       #pragma unroll
       for (int n = 0; n < 4; ++n) {
-        #pragma unroll
-        for (int c = 0; c < 2; ++c) {
-          // V^T fragment (row d, k = kv) from the row-major V image
-          bf16x8 va = read_frag_tr(lds_v[buf], 32 * c, 16 * n, lane);
-          o_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              va, pb[c], o_acc[n], 0, 0, 0);
-        }
+        bf16x8 va0, va1;
+        read_frag_tr_2row(lds_v[buf], 0, 32, 16 * n, lane, va0, va1);
+        o_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            va0, pb[0], o_acc[n], 0, 0, 0);
+        o_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            va1, pb[1], o_acc[n], 0, 0, 0);
       }
     }
     // write-late: next tile's regs -> the other buffers, one barrier
