@@ -81,8 +81,15 @@ __device__ __forceinline__ void stage_transposed(
 // fragment read over all 8 granules -> conflict-free ds_read_b128,
 // and distinct granules for the d / d+8 write groups.
 __device__ __forceinline__ int swz_off(int d, int byte_in_row) {
+  // sel = (d>>1)*3 ^ (d>>3): within any 4 consecutive rows (one
+  // ds_read_b64_tr_b16 block) the same-parity pair differs by XOR 3
+  // -> granule sets disjoint (the previous (d>>1)^(d>>3) left rows
+  // d/d+2 sharing a granule pair: 2-way tr conflicts); 3 is coprime
+  // to 8 so 8 consecutive same-parity rows still spread over all 8
+  // granules for the b128 fragment reads, and the d vs d+32 staging
+  // groups still split via the (d>>3) term.
   return d * (2 * BN) +
-         (byte_in_row ^ ((((d >> 1) ^ (d >> 3)) & 7) << 4));
+         (byte_in_row ^ (((((d >> 1) * 3) ^ (d >> 3)) & 7) << 4));
 }
 
 // Transpose-stage a [BN, HS] row-major global tile into img[HS][BN]
