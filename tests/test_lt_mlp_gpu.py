@@ -154,3 +154,9 @@ def test_fused_mlp_autograd_matches_reference():
         scale = want.grad.abs().max().item() + 1e-6
         err = (got.grad.float() - want.grad).abs().max().item()
         assert err < 5e-2 * max(scale, 1.0), f"{name}: {err}"
+
+
+def teardown_module():
+    # don't leak the experimental opt-ins into later test modules
+    os.environ.pop("RLA_LT_MLP", None)
+    os.environ.pop("RLA_FUSED_MLP", None)
