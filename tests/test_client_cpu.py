@@ -14,7 +14,6 @@ spawned driver actor, and THAT driver spawns the training workers.
 import os
 
 import pytest
-import torch
 
 from ray_lightning_amd.runtime import ActorHandle
 

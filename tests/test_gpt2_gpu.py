@@ -79,3 +79,33 @@ def test_gpt2_grads_flash_vs_sdpa():
         scale = gs[n].abs().max().item() + 1e-6
         err = (gf[n] - gs[n]).abs().max().item()
         assert err < 6e-2 * max(scale, 1.0), f"{n}: {err} (scale {scale})"
+
+
+def test_gpt2_soak_stable_memory_and_loss():
+    """60-step soak on the full fused stack: loss keeps improving, no
+    NaN/inf, and steady-state GPU memory does not creep (allocator
+    stability across flash/fused-LN/chunked-CE/optimizer reuse)."""
+    torch.manual_seed(1)
+    cfg = _tiny_cfg()
+    model = to_bf16_training(GPT2(cfg).cuda())
+    from ray_lightning_amd.optim import FusedAdamW
+    opt = FusedAdamW(model.parameters(), lr=1e-3)
+    x = torch.randint(0, cfg.vocab_size, (4, 128), device="cuda")
+    y = torch.randint(0, cfg.vocab_size, (4, 128), device="cuda")
+    losses = []
+    mem_mark = None
+    for step in range(60):
+        _, loss = model(x, y)
+        loss.backward()
+        opt.step()
+        opt.zero_grad(set_to_none=True)
+        losses.append(float(loss))
+        if step == 20:
+            torch.cuda.synchronize()
+            mem_mark = torch.cuda.memory_allocated()
+    torch.cuda.synchronize()
+    assert all(math.isfinite(v) for v in losses)
+    assert losses[-1] < losses[0] * 0.5
+    # steady-state memory must not creep after warmup
+    growth = torch.cuda.memory_allocated() - mem_mark
+    assert growth < 16 << 20, f"memory crept {growth / 1e6:.1f} MB"
