@@ -103,23 +103,28 @@ class RayLauncher:
             else:
                 assignments.append([])
 
-        # Per-node GPU visibility union (single-node runtime: the union of
-        # every worker's devices — reference _share_cuda_visible_devices).
-        union: List[int] = sorted(
-            {g for ids in assignments for g in ids})
-        union_str = ",".join(str(g) for g in union)
+        # Optional test/deployment hook: pin workers onto named nodes
+        # ("0:nodeA,1:nodeB,..."); workers report that id from
+        # get_node_ip() so rank mapping and visibility grouping see a
+        # multi-node topology.
+        node_of: Dict[int, str] = {}
+        raw_map = os.environ.get("RLA_NODE_OF_WORKER", "")
+        for part in raw_map.split(","):
+            if ":" in part:
+                k, v = part.split(":", 1)
+                node_of[int(k)] = v
 
         actor_cls = get_executable_cls() or ActorHandle
         self._workers = []
         for i in range(num_workers):
             env = dict(base_env)
             env["RLA_GPU_IDS"] = ",".join(str(g) for g in assignments[i])
-            if union:
-                env["HIP_VISIBLE_DEVICES"] = union_str
-                env["CUDA_VISIBLE_DEVICES"] = union_str
+            if i in node_of:
+                env["RLA_NODE_IP"] = node_of[i]
             worker = actor_cls(env, name=f"rla-worker-{i}")
             worker.assigned_gpu_ids = assignments[i]
             self._workers.append(worker)
+        self._assignments = assignments
 
         # init_hook runs on every worker before training
         # (reference ray_launcher.py:79-83)
@@ -136,6 +141,7 @@ class RayLauncher:
         self._setup_env_vars()
 
         self._global_to_local = self.get_local_ranks()
+        self._share_visible_devices()
 
         self.tune_queue = None
         if tune_enabled:
@@ -154,6 +160,29 @@ class RayLauncher:
             if key in os.environ:
                 env[key] = os.environ[key]
         futs = [w.set_env_vars(env) for w in self._workers]
+        for f in futs:
+            f.get(timeout=60)
+
+    def _share_visible_devices(self) -> None:
+        """PER-NODE GPU-visibility union, shipped post-spawn via the
+        env RPC before any worker touches the HIP runtime (reference
+        ray_launcher.py:177-219 _share_cuda_visible_devices — the
+        reference also groups by node ip; a single global union would
+        name devices that do not exist on a worker's node)."""
+        if not any(self._assignments):
+            return
+        infos = [f.get(timeout=60)
+                 for f in [w.get_node_and_gpu_ids()
+                           for w in self._workers]]
+        per_node: Dict[str, List[int]] = {}
+        for (ip, _), ids in zip(infos, self._assignments):
+            per_node.setdefault(ip, []).extend(ids)
+        futs = []
+        for (ip, _), w in zip(infos, self._workers):
+            union = sorted(set(per_node[ip]))
+            vis = ",".join(str(g) for g in union)
+            futs.append(w.set_env_vars({"HIP_VISIBLE_DEVICES": vis,
+                                        "CUDA_VISIBLE_DEVICES": vis}))
         for f in futs:
             f.get(timeout=60)
 

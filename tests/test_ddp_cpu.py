@@ -405,3 +405,39 @@ def test_train_four_workers_grad_agreement(tmp_path):
     # CHANGED weights and the 4-worker run completed the protocol.
     assert torch.isfinite(results[4]).all()
     assert not torch.equal(results[1], results[4]) or True
+
+
+def test_per_node_visibility_union(monkeypatch):
+    """Multi-node grouping: workers pinned onto two fake nodes must get
+    PER-NODE visibility unions and a per-node local-rank map — a global
+    union would name devices that do not exist on the other node
+    (reference _share_cuda_visible_devices groups by node ip)."""
+    from ray_lightning_amd.launchers.ray_launcher import RayLauncher
+    from ray_lightning_amd.runtime import actor as actor_mod
+
+    monkeypatch.setenv("RLA_NODE_OF_WORKER", "0:nodeA,1:nodeA,2:nodeB,3:nodeB")
+    strategy = RayStrategy(num_workers=4)
+    # fake gpu assignments: two per node
+    launcher = RayLauncher(strategy)
+    launcher.setup_workers(tune_enabled=False)
+    try:
+        # inject fake assignments and re-share (CPU container has none)
+        launcher._assignments = [[0], [1], [0], [2]]
+        launcher._share_visible_devices()
+
+        def read_env(keys):
+            import os
+            return {k: os.environ.get(k) for k in keys}
+
+        envs = [w.execute(read_env,
+                          ["HIP_VISIBLE_DEVICES", "RLA_NODE_IP"]).get(
+                              timeout=60)
+                for w in launcher._workers]
+        assert envs[0]["HIP_VISIBLE_DEVICES"] == "0,1"
+        assert envs[1]["HIP_VISIBLE_DEVICES"] == "0,1"
+        assert envs[2]["HIP_VISIBLE_DEVICES"] == "0,2"
+        assert envs[3]["HIP_VISIBLE_DEVICES"] == "0,2"
+        assert launcher._global_to_local == [(0, 0), (1, 0), (0, 1),
+                                             (1, 1)]
+    finally:
+        launcher.teardown_workers()
