@@ -992,7 +992,10 @@ __global__ __launch_bounds__(256, 4) void flash_fwd_v3_kernel(
     const bool active = kn0 <= q0 + WQ - 1;
     if (active) {
       // --- S^T = K Q^T: lane gets kv = kn0+16*sub+4*g+r of column myq
+      // (T5: favor the MFMA-entering wave over waves in their
+      // memory/softmax phases)
       f32x4 st[4];
+      __builtin_amdgcn_s_setprio(1);
       #pragma unroll
       for (int sub = 0; sub < 4; ++sub) {
         bf16x8 kf0 = read_frag_swz(lds_k[buf], 16 * sub, 0, lane);
@@ -1002,6 +1005,7 @@ __global__ __launch_bounds__(256, 4) void flash_fwd_v3_kernel(
         st[sub] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             kf1, qf[1], st[sub], 0, 0, 0);
       }
+      __builtin_amdgcn_s_setprio(0);
       // --- causal mask (only tiles overlapping the diagonal) ---------
       if (kn0 + BN - 1 > q0) {
         #pragma unroll
@@ -1071,6 +1075,7 @@ __global__ __launch_bounds__(256, 4) void flash_fwd_v3_kernel(
       }
       // --- O^T += V^T P^T: both kv chunks' V^T fragments per
       // d-block with one drain
+      __builtin_amdgcn_s_setprio(1);
       #pragma unroll
       for (int n = 0; n < 4; ++n) {
         bf16x8 va0, va1;
@@ -1080,6 +1085,7 @@ __global__ __launch_bounds__(256, 4) void flash_fwd_v3_kernel(
         o_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             va1, pb[1], o_acc[n], 0, 0, 0);
       }
+      __builtin_amdgcn_s_setprio(0);
     }
     // write-late: next tile's regs -> the other buffers, one barrier
     stage_r_write(lds_v[buf ^ 1], vt);
@@ -1293,6 +1299,7 @@ __global__ __launch_bounds__(256, 3) void flash_bwd_dkv_v3_kernel(
         bf16x8 aq1 = read_frag_swz(lds_q[buf], 16 * sub, 1, lane);
         bf16x8 ad0 = read_frag_swz(lds_do[buf], 16 * sub, 0, lane);
         bf16x8 ad1 = read_frag_swz(lds_do[buf], 16 * sub, 1, lane);
+        __builtin_amdgcn_s_setprio(1);
         f32x4 sacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             aq0, kf[0], f32x4{0.f, 0.f, 0.f, 0.f}, 0, 0, 0);
         sacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -1301,6 +1308,7 @@ __global__ __launch_bounds__(256, 3) void flash_bwd_dkv_v3_kernel(
             ad0, vf[0], f32x4{0.f, 0.f, 0.f, 0.f}, 0, 0, 0);
         dpacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             ad1, vf[1], dpacc, 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
         #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int qi = 16 * sub + 4 * g + r;     // tile-local q
@@ -1345,6 +1353,7 @@ __global__ __launch_bounds__(256, 3) void flash_bwd_dkv_v3_kernel(
           pw[0] = d0; pw[1] = d1; pw[2] = d2; pw[3] = d3;
         }
         // d-blocks in pairs: dO^T / Q^T fragments, one drain per pair
+        __builtin_amdgcn_s_setprio(1);
         #pragma unroll
         for (int np = 0; np < 2; ++np) {
           bf16x8 adot[2], aqt[2];
@@ -1358,6 +1367,7 @@ __global__ __launch_bounds__(256, 3) void flash_bwd_dkv_v3_kernel(
                 aqt[n], db, dk_acc[2 * np + n], 0, 0, 0);
           }
         }
+        __builtin_amdgcn_s_setprio(0);
       }
     }
     // write-late: next tile -> other buffers, one barrier per tile
