@@ -109,7 +109,13 @@ def _worker(rank: int, port: int, mode: str, out_q) -> None:
         elif mode == "sharded_state":
             from ray_lightning_amd.engine.sharded import (ShardedDDP,
                                                           ShardedOptimizer)
-            opt = torch.optim.Adam(model.parameters(), lr=0.01)
+            # TWO param groups with distinct hyperparameters: the
+            # consolidated dict must partition global indices per group
+            # (fairscale/torch format — the r01 ADVICE finding)
+            params = list(model.parameters())
+            opt = torch.optim.Adam(
+                [{"params": params[:2], "weight_decay": 0.1},
+                 {"params": params[2:], "weight_decay": 0.0}], lr=0.01)
             oss = ShardedOptimizer(opt, comm, bucket_cap_mb=0.0001)
             sddp = ShardedDDP(model, comm, oss, bucket_cap_mb=0.0001)
             for _step in range(3):
@@ -123,7 +129,11 @@ def _worker(rank: int, port: int, mode: str, out_q) -> None:
                 out = {gi: {k: (v.numpy() if torch.is_tensor(v) else v)
                             for k, v in st.items()}
                        for gi, st in state.items()}
-                out_q.put((rank, "ok", out))
+                groups = [{k: (list(v) if k == "params" else v)
+                           for k, v in g.items()}
+                          for g in sd["consolidated"]["param_groups"]]
+                out_q.put((rank, "ok", {"state": out,
+                                        "param_groups": groups}))
             else:
                 out_q.put((rank, "ok", None))
         elif mode == "sharded":
@@ -228,11 +238,40 @@ def test_sharded_params_match_plain_sgd():
 
 def test_sharded_optimizer_state_consolidation():
     """Consolidated sharded-Adam state == single-process Adam state on
-    the averaged gradients (every param present, exp_avg equal)."""
+    the averaged gradients (every param present, exp_avg equal), and
+    the consolidated dict loads into a PLAIN torch optimizer with the
+    per-group hyperparameters intact (the r01 ADVICE scenario)."""
     results = _run_workers("sharded_state")
-    state = results[0]
+    payload = results[0]
+    state = payload["state"]
+    groups = payload["param_groups"]
+    # group partition: disjoint global indices, correct hypers
+    n_params = 4  # Linear(4,8)+Linear(8,2) -> 4 tensors
+    assert sorted(groups[0]["params"] + groups[1]["params"]) == \
+        list(range(n_params))
+    assert groups[0]["weight_decay"] == 0.1
+    assert groups[1]["weight_decay"] == 0.0
+    # load into a plain torch optimizer (standard format consumer)
+    plain_model = _make_model()
+    pp = list(plain_model.parameters())
+    plain = torch.optim.Adam(
+        [{"params": pp[:2], "weight_decay": 0.1},
+         {"params": pp[2:], "weight_decay": 0.0}], lr=0.01)
+    plain.load_state_dict({
+        "state": {int(k): {kk: (torch.tensor(vv)
+                                if not torch.is_tensor(vv) and
+                                not isinstance(vv, (int, float))
+                                else vv)
+                           for kk, vv in st.items()}
+                  for k, st in state.items()},
+        "param_groups": groups})
+    assert plain.param_groups[0]["weight_decay"] == 0.1
+
     model = _make_model()
-    opt = torch.optim.Adam(model.parameters(), lr=0.01)
+    mp = list(model.parameters())
+    opt = torch.optim.Adam(
+        [{"params": mp[:2], "weight_decay": 0.1},
+         {"params": mp[2:], "weight_decay": 0.0}], lr=0.01)
     for _step in range(3):
         opt.zero_grad()
         losses = [model(_rank_batch(r)).pow(2).mean()
