@@ -92,65 +92,11 @@ __device__ __forceinline__ int swz_off(int d, int byte_in_row) {
          (byte_in_row ^ (((((d >> 1) * 3) ^ (d >> 3)) & 7) << 4));
 }
 
-// Transpose-stage a [BN, HS] row-major global tile into img[HS][BN]
-// with dword-paired writes (two rows per ds_write_b32: 8 b32 writes
-// per thread at <=2-way instead of 16 scalar b16 at up to 8-way).
-__device__ __forceinline__ void stage_transposed_swz(
-    __bf16* dst, const __hip_bfloat16* src, int row0) {
-  const int p = threadIdx.x >> 3;        // row pair: rows 2p, 2p+1
-  const int c0 = (threadIdx.x & 7) * 8;  // 8-column run (8 consecutive
-  // lanes cover one 128 B source row: coalesced global loads)
-  bf16x8 t0, t1;
-  *reinterpret_cast<int4*>(&t0) = *reinterpret_cast<const int4*>(
-      src + (long)(row0 + 2 * p) * HS + c0);
-  *reinterpret_cast<int4*>(&t1) = *reinterpret_cast<const int4*>(
-      src + (long)(row0 + 2 * p + 1) * HS + c0);
-  char* base = reinterpret_cast<char*>(dst);
-  #pragma unroll
-  for (int i = 0; i < 8; ++i) {
-    union { unsigned u; __bf16 h[2]; } w;
-    w.h[0] = t0[i];
-    w.h[1] = t1[i];
-    *reinterpret_cast<unsigned*>(
-        base + swz_off(c0 + i, 4 * p)) = w.u;
-  }
-}
-
 // Split staging (T14 issue-early/write-late): load the next tile's
 // piece into registers at the top of the current tile's compute, write
 // it to the other LDS buffer after the compute — HBM latency hides
 // under the MFMA/softmax phase instead of sitting in front of the
 // barrier.
-struct StageRegsT {   // V^T transpose staging piece (2 source rows)
-  bf16x8 r0, r1;
-};
-
-__device__ __forceinline__ StageRegsT stage_t_load(
-    const __hip_bfloat16* src, int row0) {
-  const int p = threadIdx.x >> 3;
-  const int c0 = (threadIdx.x & 7) * 8;
-  StageRegsT t;
-  *reinterpret_cast<int4*>(&t.r0) = *reinterpret_cast<const int4*>(
-      src + (long)(row0 + 2 * p) * HS + c0);
-  *reinterpret_cast<int4*>(&t.r1) = *reinterpret_cast<const int4*>(
-      src + (long)(row0 + 2 * p + 1) * HS + c0);
-  return t;
-}
-
-__device__ __forceinline__ void stage_t_write(__bf16* dst,
-                                              const StageRegsT& t) {
-  const int p = threadIdx.x >> 3;
-  const int c0 = (threadIdx.x & 7) * 8;
-  char* base = reinterpret_cast<char*>(dst);
-  #pragma unroll
-  for (int i = 0; i < 8; ++i) {
-    union { unsigned u; __bf16 h[2]; } w;
-    w.h[0] = t.r0[i];
-    w.h[1] = t.r1[i];
-    *reinterpret_cast<unsigned*>(base + swz_off(c0 + i, 4 * p)) = w.u;
-  }
-}
-
 struct StageRegsR {   // row-major staging piece (rows r, r+32)
   bf16x8 r0, r1;
 };
@@ -262,44 +208,6 @@ __device__ __forceinline__ void read_frag_tr_2row(const __bf16* img,
   wa[0] = ra0[0]; wa[1] = ra0[1]; wa[2] = ra1[0]; wa[3] = ra1[1];
   unsigned* wb = reinterpret_cast<unsigned*>(&vb);
   wb[0] = rb0[0]; wb[1] = rb0[1]; wb[2] = rb1[0]; wb[3] = rb1[1];
-}
-
-// Paired variant: transposed fragments from TWO images with a single
-// lgkm drain (dkv reads dO^T and Q^T back to back — separate helper
-// calls would serialize on their internal waits).
-__device__ __forceinline__ void read_frag_tr2(const __bf16* imgA,
-                                              const __bf16* imgB,
-                                              int krow0, int col16,
-                                              int lane, bf16x8& va,
-                                              bf16x8& vb) {
-  const int j = lane & 15;
-  const int g = lane >> 4;
-  const unsigned baseA =
-      (unsigned)(size_t)(__attribute__((address_space(3))) const char*)
-          (const void*)imgA;
-  const unsigned baseB =
-      (unsigned)(size_t)(__attribute__((address_space(3))) const char*)
-          (const void*)imgB;
-  const int r0 = krow0 + 8 * g + (j >> 2);
-  const int cb = (col16 + 4 * (j & 3)) * 2;
-  const unsigned o0 = (unsigned)swz_off(r0, cb);
-  const unsigned o1 = (unsigned)swz_off(r0 + 4, cb);
-  typedef __attribute__((ext_vector_type(2))) unsigned uint2v;
-  uint2v a0, a1, b0, b1;
-  asm volatile(
-      "ds_read_b64_tr_b16 %0, %4\n"
-      "ds_read_b64_tr_b16 %1, %5\n"
-      "ds_read_b64_tr_b16 %2, %6\n"
-      "ds_read_b64_tr_b16 %3, %7\n"
-      "s_waitcnt lgkmcnt(0)"
-      : "=&v"(a0), "=&v"(a1), "=&v"(b0), "=&v"(b1)
-      : "v"(baseA + o0), "v"(baseA + o1), "v"(baseB + o0),
-        "v"(baseB + o1)
-      : "memory");
-  unsigned* wa = reinterpret_cast<unsigned*>(&va);
-  wa[0] = a0[0]; wa[1] = a0[1]; wa[2] = a1[0]; wa[3] = a1[1];
-  unsigned* wb = reinterpret_cast<unsigned*>(&vb);
-  wb[0] = b0[0]; wb[1] = b0[1]; wb[2] = b1[0]; wb[3] = b1[1];
 }
 
 // ---------------------------------------------------------------------
@@ -1615,35 +1523,6 @@ __device__ __forceinline__ bf16x8 read_frag_swz32(const __bf16* img,
   *reinterpret_cast<int4*>(&v) = *reinterpret_cast<const int4*>(
       reinterpret_cast<const char*>(img) +
       swz_off(row, 32 * c + 16 * (lane >> 5)));
-  return v;
-}
-
-// Transposed A-fragment via ds_read_b64_tr_b16 with explicit row/col
-// bases (tr16_probe semantics: 16-lane group -> [4 src-rows][16 col]
-// block, lane j of the group receives column j).
-__device__ __forceinline__ bf16x8 read_frag_tr4(const __bf16* img,
-                                                int krow0, int col16,
-                                                int lane) {
-  const int j = lane & 15;
-  const unsigned base =
-      (unsigned)(size_t)(__attribute__((address_space(3))) const char*)
-          (const void*)img;
-  const int r0 = krow0 + (j >> 2);
-  const int cb = (col16 + 4 * (j & 3)) * 2;
-  const unsigned a0 = base + (unsigned)swz_off(r0, cb);
-  const unsigned a1 = base + (unsigned)swz_off(r0 + 4, cb);
-  typedef __attribute__((ext_vector_type(2))) unsigned uint2v;
-  uint2v lo, hi;
-  asm volatile(
-      "ds_read_b64_tr_b16 %0, %2\n"
-      "ds_read_b64_tr_b16 %1, %3\n"
-      "s_waitcnt lgkmcnt(0)"
-      : "=&v"(lo), "=&v"(hi)
-      : "v"(a0), "v"(a1)
-      : "memory");
-  bf16x8 v;
-  unsigned* w = reinterpret_cast<unsigned*>(&v);
-  w[0] = lo[0]; w[1] = lo[1]; w[2] = hi[0]; w[3] = hi[1];
   return v;
 }
 
