@@ -27,6 +27,15 @@ constexpr int BM = 128;     // query rows per workgroup (4 waves)
 constexpr int BN = 64;      // key/value rows per tile
 constexpr float kNegInf = -1e30f;
 
+// Raw v_exp_f32 (2^x): libm exp2f carries ~5 range-guard instructions
+// per call (ldexp/cmp/cndmask) — the fwd softmax issues 17 of them per
+// tile and the guards are dead weight here (inputs are <= 8 by the
+// defer-max bound and masked entries sit at ~-1.8e29, which the raw
+// instruction correctly underflows to 0).
+__device__ __forceinline__ float fast_exp2(float x) {
+  return __builtin_amdgcn_exp2f(x);
+}
+
 __device__ __forceinline__ float row_reduce_max(float v) {
   #pragma unroll
   for (int w = 8; w >= 1; w >>= 1)
@@ -530,7 +539,7 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
             const int kvrow = wkv0 + mh * 16 + (lane >> 4) * 4 + r;
             const bool valid = (qm0 + qcol) >= kvrow;
             const float p = valid
-                ? exp2f((scale * st[r] - lse_q) * l2e) : 0.f;
+                ? fast_exp2((scale * st[r] - lse_q) * l2e) : 0.f;
             const float ds = valid
                 ? scale * p * (dpt[r] - d_q) : 0.f;
             pt[(mh * 16 + (lane >> 4) * 4 + r) * BN + qcol] =
@@ -944,7 +953,7 @@ __global__ __launch_bounds__(256, 4) void flash_fwd_v3_kernel(
         corr = 1.0f;
       } else {
         m_new = fmaxf(m_i, mx);
-        corr = exp2f((m_i - m_new) * l2e);
+        corr = fast_exp2((m_i - m_new) * l2e);
         #pragma unroll
         for (int n = 0; n < 4; ++n)
           #pragma unroll
@@ -956,8 +965,10 @@ __global__ __launch_bounds__(256, 4) void flash_fwd_v3_kernel(
       for (int sub = 0; sub < 4; ++sub)
         #pragma unroll
         for (int r = 0; r < 4; ++r) {
-          const float p = (st[sub][r] <= kNegInf * 0.5f)
-              ? 0.f : exp2f((st[sub][r] - m_new) * l2e);
+          // masked entries are kNegInf: the raw exp2 underflows the
+          // ~-1.8e29 exponent to exactly 0 — no guard needed (every
+          // lane's FIRST tile contains kv<=q, so m_new is finite)
+          const float p = fast_exp2((st[sub][r] - m_new) * l2e);
           st[sub][r] = p;
           rowsum += p;
         }
@@ -1222,7 +1233,8 @@ __global__ __launch_bounds__(256, 3) void flash_bwd_dkv_v3_kernel(
           const int qi = 16 * sub + 4 * g + r;     // tile-local q
           const bool valid = (qm0 + qi) >= mykv;
           const float p = valid
-              ? exp2f((scale * sacc[r] - lds_lse[buf][qi]) * l2e) : 0.f;
+              ? fast_exp2((scale * sacc[r] - lds_lse[buf][qi]) * l2e)
+              : 0.f;
           pv[sub][r] = p;
           dsv[sub][r] = valid
               ? scale * p * (dpacc[r] - lds_d[buf][qi]) : 0.f;
@@ -1396,7 +1408,7 @@ __global__ __launch_bounds__(256, 4) void flash_bwd_dq_v3_kernel(
           const int kvr = kn0 + 16 * sub + 4 * g + r;
           const bool valid = kvr <= myq;
           const float p = valid
-              ? exp2f((scale * st[r] - lse_q) * l2e) : 0.f;
+              ? fast_exp2((scale * st[r] - lse_q) * l2e) : 0.f;
           dsv[sub][r] = valid
               ? scale * p * (dpt[r] - d_q) : 0.f;
         }
@@ -1655,7 +1667,7 @@ __global__ __launch_bounds__(256) void flash_fwd_v4_kernel(
         corr = 1.0f;
       } else {
         m_new = fmaxf(m_i, mx);
-        corr = exp2f((m_i - m_new) * l2e);
+        corr = fast_exp2((m_i - m_new) * l2e);
         #pragma unroll
         for (int dt = 0; dt < 2; ++dt)
           #pragma unroll
@@ -1667,8 +1679,7 @@ __global__ __launch_bounds__(256) void flash_fwd_v4_kernel(
       for (int sub = 0; sub < 2; ++sub)
         #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          const float p = (st[sub][r] <= kNegInf * 0.5f)
-              ? 0.f : exp2f((st[sub][r] - m_new) * l2e);
+          const float p = fast_exp2((st[sub][r] - m_new) * l2e);
           st[sub][r] = p;
           rowsum += p;
         }
