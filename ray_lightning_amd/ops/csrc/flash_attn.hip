@@ -1228,16 +1228,24 @@ __global__ __launch_bounds__(256, 3) void flash_bwd_dkv_v3_kernel(
         dpacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             ad1, vf[1], dpacc, 0, 0, 0);
         __builtin_amdgcn_s_setprio(0);
+        // lse/D for this lane's 4 consecutive q rows: one b128 each
+        // instead of 8 scalar ds_read_b32 (the bwd census showed the
+        // per-element reads + their waits dominating non-MFMA issue)
+        const float4 lse4 = *reinterpret_cast<const float4*>(
+            &lds_lse[buf][16 * sub + 4 * g]);
+        const float4 d4 = *reinterpret_cast<const float4*>(
+            &lds_d[buf][16 * sub + 4 * g]);
+        const float lse_a[4] = {lse4.x, lse4.y, lse4.z, lse4.w};
+        const float d_a[4] = {d4.x, d4.y, d4.z, d4.w};
         #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int qi = 16 * sub + 4 * g + r;     // tile-local q
           const bool valid = (qm0 + qi) >= mykv;
           const float p = valid
-              ? fast_exp2((scale * sacc[r] - lds_lse[buf][qi]) * l2e)
-              : 0.f;
+              ? fast_exp2((scale * sacc[r] - lse_a[r]) * l2e) : 0.f;
           pv[sub][r] = p;
           dsv[sub][r] = valid
-              ? scale * p * (dpacc[r] - lds_d[buf][qi]) : 0.f;
+              ? scale * p * (dpacc[r] - d_a[r]) : 0.f;
         }
       }
       // redistribute over the q axis -> B-fragments (k = q)
