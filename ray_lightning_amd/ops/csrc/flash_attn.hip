@@ -1333,6 +1333,42 @@ __global__ __launch_bounds__(256, 3) void flash_bwd_dkv_v3_kernel(
   }
 }
 
+// Column-paired tr reads for dq: two 16-col blocks of the SAME rows
+// from one image, single lgkm drain.
+__device__ __forceinline__ void read_frag_tr_2col(const __bf16* img,
+                                                  int krow0, int col16a,
+                                                  int col16b, int lane,
+                                                  bf16x8& va,
+                                                  bf16x8& vb) {
+  const int j = lane & 15;
+  const int g = lane >> 4;
+  const unsigned base =
+      (unsigned)(size_t)(__attribute__((address_space(3))) const char*)
+          (const void*)img;
+  const int r0 = krow0 + 8 * g + (j >> 2);
+  const int cba = (col16a + 4 * (j & 3)) * 2;
+  const int cbb = (col16b + 4 * (j & 3)) * 2;
+  const unsigned a0 = base + (unsigned)swz_off(r0, cba);
+  const unsigned a1 = base + (unsigned)swz_off(r0 + 4, cba);
+  const unsigned b0 = base + (unsigned)swz_off(r0, cbb);
+  const unsigned b1 = base + (unsigned)swz_off(r0 + 4, cbb);
+  typedef __attribute__((ext_vector_type(2))) unsigned uint2v;
+  uint2v ra0, ra1, rb0, rb1;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %4\n"
+      "ds_read_b64_tr_b16 %1, %5\n"
+      "ds_read_b64_tr_b16 %2, %6\n"
+      "ds_read_b64_tr_b16 %3, %7\n"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(ra0), "=&v"(ra1), "=&v"(rb0), "=&v"(rb1)
+      : "v"(a0), "v"(a1), "v"(b0), "v"(b1)
+      : "memory");
+  unsigned* wa = reinterpret_cast<unsigned*>(&va);
+  wa[0] = ra0[0]; wa[1] = ra0[1]; wa[2] = ra1[0]; wa[3] = ra1[1];
+  unsigned* wb = reinterpret_cast<unsigned*>(&vb);
+  wb[0] = rb0[0]; wb[1] = rb0[1]; wb[2] = rb1[0]; wb[3] = rb1[1];
+}
+
 // dQ v3: workgroup = 64 q rows, 4 waves x 16 q columns; iterate kv
 // tiles of 64. lse/D are lane-resident (q fixed per lane).
 //   S^T[kv][q]  = mfma(K_frag,  Q^T B-frag)
@@ -1435,13 +1471,17 @@ __global__ __launch_bounds__(256, 4) void flash_bwd_dq_v3_kernel(
         redist_pair<USE_PERMLANE>(x1, y1, lane, d1, d3);
         unsigned* pw = reinterpret_cast<unsigned*>(&db);
         pw[0] = d0; pw[1] = d1; pw[2] = d2; pw[3] = d3;
+        // K^T fragments (row d, k = kv) in column pairs: one drain
+        // per pair instead of one per d-block
         #pragma unroll
-        for (int n = 0; n < 4; ++n) {
-          // K^T fragment (row d, k = kv) via hardware-transpose read
-          bf16x8 akt = read_frag_tr(lds_kk[buf], 32 * blk, 16 * n,
-                                    lane);
-          dq_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              akt, db, dq_acc[n], 0, 0, 0);
+        for (int np = 0; np < 2; ++np) {
+          bf16x8 ak0, ak1;
+          read_frag_tr_2col(lds_kk[buf], 32 * blk, 32 * np,
+                            32 * np + 16, lane, ak0, ak1);
+          dq_acc[2 * np] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              ak0, db, dq_acc[2 * np], 0, 0, 0);
+          dq_acc[2 * np + 1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              ak1, db, dq_acc[2 * np + 1], 0, 0, 0);
         }
       }
     }
