@@ -1241,11 +1241,14 @@ __global__ __launch_bounds__(256, 3) void flash_bwd_dkv_v3_kernel(
         for (int r = 0; r < 4; ++r) {
           const int qi = 16 * sub + 4 * g + r;     // tile-local q
           const bool valid = (qm0 + qi) >= mykv;
-          const float p = valid
-              ? fast_exp2((scale * sacc[r] - lse_a[r]) * l2e) : 0.f;
+          // mask the EXPONENT instead of p and ds: raw exp2 of -1e30
+          // is exactly 0, and ds = scale*p*(...) inherits the zero —
+          // one select instead of two per element
+          const float e = valid
+              ? (scale * sacc[r] - lse_a[r]) * l2e : -1e30f;
+          const float p = fast_exp2(e);
           pv[sub][r] = p;
-          dsv[sub][r] = valid
-              ? scale * p * (dpacc[r] - d_a[r]) : 0.f;
+          dsv[sub][r] = scale * p * (dpacc[r] - d_a[r]);
         }
       }
       // redistribute over the q axis -> B-fragments (k = q)
@@ -1451,10 +1454,9 @@ __global__ __launch_bounds__(256, 4) void flash_bwd_dq_v3_kernel(
         for (int r = 0; r < 4; ++r) {
           const int kvr = kn0 + 16 * sub + 4 * g + r;
           const bool valid = kvr <= myq;
-          const float p = valid
-              ? fast_exp2((scale * st[r] - lse_q) * l2e) : 0.f;
-          dsv[sub][r] = valid
-              ? scale * p * (dpt[r] - d_q) : 0.f;
+          const float e = valid
+              ? (scale * st[r] - lse_q) * l2e : -1e30f;
+          dsv[sub][r] = scale * fast_exp2(e) * (dpt[r] - d_q);
         }
       }
       #pragma unroll
