@@ -4,9 +4,9 @@
 The custom v3 kernels (swapped-operand S^T, in-register permlane
 redistribution, ds_read_b64_tr_b16 transposed fragments — see
 flash_attn.hip) engage for causal bf16 hs=64 T%64==0 on GPU and are
-the DEFAULT: measured 0.463 ms fwd+bwd vs AOTriton SDPA's 0.715 ms on
-the GPT-2-XL shape (B=8, H=25, T=1024) — 35% faster, winning both
-directions; profiles/r02_flash_v3.md.
+the DEFAULT: measured 0.421 ms fwd+bwd vs AOTriton SDPA's 0.710 ms on
+the GPT-2-XL shape (B=8, H=25, T=1024) — 41% faster, winning both
+directions (fwd -12%, bwd -47%); profiles/r02_flash_v3.md.
 Anything else falls back to ``F.scaled_dot_product_attention``.
 ``RLA_FLASH=0`` disables the custom path; ``RLA_FLASH_SHFL=1`` selects
 the ds_bpermute redistribution variant (debug).
