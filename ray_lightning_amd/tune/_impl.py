@@ -103,16 +103,23 @@ def _expand_configs(space: Dict[str, Any], num_samples: int,
 def get_tune_resources(num_workers: int = 1,
                        num_cpus_per_worker: int = 1,
                        use_gpu: bool = False,
-                       cpus_per_worker: Optional[int] = None
+                       cpus_per_worker: Optional[int] = None,
+                       resources_per_worker: Optional[Dict] = None
                        ) -> PlacementGroup:
-    """PACK placement group: 1 head CPU + per-worker bundles."""
+    """PACK placement group: 1 head CPU + per-worker bundles.
+
+    ``resources_per_worker`` mirrors the strategy ctor (e.g.
+    ``{"GPU": 0.25}`` for fractional-GPU trials)."""
     if cpus_per_worker is not None:
         import warnings
         warnings.warn("`cpus_per_worker` is deprecated; use "
                       "`num_cpus_per_worker`.", PendingDeprecationWarning)
         num_cpus_per_worker = cpus_per_worker
+    rpw = dict(resources_per_worker or {})
+    gpu_demand = rpw.pop("GPU", int(use_gpu))
+    cpu_demand = rpw.pop("CPU", num_cpus_per_worker)
     head_bundle = {"CPU": 1}
-    child_bundle = {"CPU": num_cpus_per_worker, "GPU": int(use_gpu)}
+    child_bundle = {"CPU": cpu_demand, "GPU": gpu_demand}
     bundles = [head_bundle] + [dict(child_bundle)
                                for _ in range(num_workers)]
     return PlacementGroup(bundles, strategy="PACK")

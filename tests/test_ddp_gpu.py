@@ -107,3 +107,42 @@ def test_tune_trial_on_gpu(tmp_path):
                                                use_gpu=True),
         local_dir=str(tmp_path / "tune"), metric="loss", mode="min")
     assert analysis.trials[0].last_result["training_iteration"] == 1
+
+
+def test_tune_two_concurrent_fractional_gpu_trials(tmp_path):
+    """TWO Tune trials concurrently, each a 2-worker RayStrategy at
+    GPU=0.25/worker, all sharing one MI355X (BASELINE config 5's
+    trial-concurrency shape downsized to one device; gloo data plane
+    for the fractional workers)."""
+    import time
+
+    from ray_lightning_amd import tune
+    from ray_lightning_amd.tune import (TuneReportCallback,
+                                        get_tune_resources)
+
+    def train_fn(config):
+        model = BoringModel()
+        trainer = Trainer(
+            default_root_dir=config["root"], max_epochs=1,
+            strategy=RayStrategy(
+                num_workers=2,
+                resources_per_worker={"GPU": 0.25}),
+            callbacks=[TuneReportCallback({"loss": "x"},
+                                          on="validation_end")],
+            limit_train_batches=4, limit_val_batches=2,
+            num_sanity_val_steps=0, enable_checkpointing=False)
+        trainer.fit(model)
+
+    t0 = time.time()
+    analysis = tune.run(
+        train_fn,
+        config={"root": str(tmp_path),
+                "trial": tune.grid_search([0, 1])},
+        resources_per_trial=get_tune_resources(
+            num_workers=2, use_gpu=True,
+            resources_per_worker={"GPU": 0.25}),
+        local_dir=str(tmp_path / "tune"), metric="loss", mode="min")
+    assert len(analysis.trials) == 2
+    for t in analysis.trials:
+        assert t.last_result["training_iteration"] == 1
+    assert time.time() - t0 < 300
