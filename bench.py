@@ -252,8 +252,10 @@ def build_comm(rank: int, world: int, device: torch.device):
     control = TorchDistCommunicator()
     # PL_TORCH_DISTRIBUTED_BACKEND=gloo forces the CPU data plane (the
     # fractional-GPU / ranks-sharing-one-device case, like the strategy)
+    local_world = int(os.environ.get("LOCAL_WORLD_SIZE", str(world)))
     if device.type == "cuda" and \
-            os.environ.get("PL_TORCH_DISTRIBUTED_BACKEND") != "gloo":
+            os.environ.get("PL_TORCH_DISTRIBUTED_BACKEND") != "gloo" \
+            and local_world <= torch.cuda.device_count():
         from ray_lightning_amd.engine.rccl import (NativeRcclCommunicator,
                                                    rccl_available)
         if rccl_available():

@@ -209,6 +209,14 @@ class RayStrategy(Strategy):
             return
         backend_override = os.environ.get("PL_TORCH_DISTRIBUTED_BACKEND")
         whole_gpus = self.num_gpus_per_worker >= 1 or self._external_mode
+        if self._external_mode and torch.cuda.is_available():
+            # ranks sharing one device (validation boxes with fewer
+            # GPUs than ranks) cannot run RCCL — fall back to gloo
+            # instead of hanging in ncclCommInitRank
+            local_world = int(os.environ.get(
+                "LOCAL_WORLD_SIZE", os.environ.get("WORLD_SIZE", "1")))
+            if local_world > torch.cuda.device_count():
+                whole_gpus = False
         if self.use_gpu and whole_gpus and torch.cuda.is_available() \
                 and backend_override != "gloo":
             from ..engine.rccl import NativeRcclCommunicator, rccl_available
