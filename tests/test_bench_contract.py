@@ -81,3 +81,23 @@ def test_bench_gpt2_sharded_labels():
     _check_contract(j, 1)
     assert "sharded" in j["config"]["parallelism"]
     assert "tokens_per_s" in j["config"]
+
+
+def test_sweep_harness_cpu():
+    """scripts/sweep_buckets.py produces a table (guards the 8-GPU
+    tuning harness the first multi-GPU node will use)."""
+    out_md = os.path.join(REPO, "profiles", "_sweep_test_tmp.md")
+    try:
+        r = subprocess.run(
+            [sys.executable, os.path.join(REPO, "scripts",
+                                          "sweep_buckets.py"),
+             "--gpus", "1", "--steps", "1", "--warmup", "0",
+             "--batch-size", "4", "--buckets", "25", "--dtypes",
+             "none", "--out", out_md],
+            capture_output=True, text=True, timeout=420, cwd=REPO)
+        assert r.returncode == 0, r.stderr[-1500:]
+        body = open(out_md).read()
+        assert "| bucket MB |" in body and "25.0" in body
+    finally:
+        if os.path.exists(out_md):
+            os.remove(out_md)

@@ -146,3 +146,19 @@ def test_tune_two_concurrent_fractional_gpu_trials(tmp_path):
     for t in analysis.trials:
         assert t.last_result["training_iteration"] == 1
     assert time.time() - t0 < 300
+
+
+def test_horovod_strategy_on_gpu(tmp_path):
+    """HorovodRayStrategy trains on a GPU worker (the third strategy's
+    GPU tier; allreduce routes through the same engine)."""
+    from ray_lightning_amd.strategies.ray_horovod import \
+        HorovodRayStrategy
+
+    model = BoringModel()
+    trainer = Trainer(
+        default_root_dir=str(tmp_path), max_epochs=1,
+        strategy=HorovodRayStrategy(num_workers=1, use_gpu=True),
+        limit_train_batches=4, limit_val_batches=2,
+        num_sanity_val_steps=0, enable_checkpointing=False)
+    trainer.fit(model)
+    assert trainer.state.finished
