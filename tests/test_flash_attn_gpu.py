@@ -157,7 +157,7 @@ def test_perm_redistribution_probe():
 
 @pytest.mark.parametrize("use_permlane", [False, True])
 @pytest.mark.parametrize("B,H,T", [(2, 3, 256), (1, 2, 1024),
-                                   (1, 1, 64)])
+                                   (1, 1, 64), (1, 2, 2048)])
 def test_flash_fwd_v3_numerics(B, H, T, use_permlane):
     """v3 (swapped-operand S^T, in-register P redistribution) against
     the fp32 reference; both redistribution paths."""
@@ -198,7 +198,7 @@ def test_flash_fwd_v3_matches_v2_lse():
 
 @pytest.mark.parametrize("use_permlane", [False, True])
 @pytest.mark.parametrize("B,H,T", [(2, 3, 256), (1, 2, 1024),
-                                   (1, 1, 64)])
+                                   (1, 1, 64), (1, 2, 2048)])
 def test_flash_bwd_v3_numerics(B, H, T, use_permlane):
     """v3 backward (swapped C layouts, in-register dS/P redistribution)
     against torch autograd through the fp32 reference."""
